@@ -1,0 +1,5 @@
+from chainermn_amd.links.batch_normalization import MultiNodeBatchNormalization  # noqa: F401
+from chainermn_amd.links.create_mnbn_model import create_mnbn_model  # noqa: F401
+from chainermn_amd.links.multi_node_chain_list import MultiNodeChainList  # noqa: F401
+from chainermn_amd.links.n_step_rnn import create_multi_node_n_step_rnn  # noqa: F401
+from chainermn_amd.links.n_step_rnn import MultiNodeNStepRNN  # noqa: F401

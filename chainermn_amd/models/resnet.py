@@ -1,0 +1,97 @@
+"""ResNet-50 — the benchmark workload (BASELINE.json configs 2/3/5).
+
+Written from the architecture (He et al. 2015, bottleneck v1 with the
+stride-on-3x3 refinement); the reference's equivalent was
+examples/imagenet/models/resnet50.py.  ``norm_layer`` is injectable so
+``create_mnbn_model`` / direct construction can use
+MultiNodeBatchNormalization for the large-batch config.
+
+channels_last memory format is the MI355X-friendly layout for MIOpen convs;
+``bench.py`` converts the module, nothing here assumes NCHW strides.
+"""
+
+import torch.nn as nn
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_ch, width, stride=1, downsample=None,
+                 norm_layer=nn.BatchNorm2d):
+        super().__init__()
+        out_ch = width * self.expansion
+        self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
+        self.bn1 = norm_layer(width)
+        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+                               bias=False)
+        self.bn2 = norm_layer(width)
+        self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
+        self.bn3 = norm_layer(out_ch)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out += identity
+        return self.relu(out)
+
+
+class ResNet50(nn.Module):
+
+    def __init__(self, num_classes=1000, norm_layer=nn.BatchNorm2d,
+                 zero_init_residual=True):
+        super().__init__()
+        self._norm = norm_layer
+        self.in_ch = 64
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = norm_layer(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_layer(64, 3, stride=1)
+        self.layer2 = self._make_layer(128, 4, stride=2)
+        self.layer3 = self._make_layer(256, 6, stride=2)
+        self.layer4 = self._make_layer(512, 3, stride=2)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode='fan_out',
+                                        nonlinearity='relu')
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+        if zero_init_residual:
+            for m in self.modules():
+                if isinstance(m, Bottleneck) and m.bn3.weight is not None:
+                    nn.init.zeros_(m.bn3.weight)
+
+    def _make_layer(self, width, blocks, stride):
+        norm_layer = self._norm
+        out_ch = width * Bottleneck.expansion
+        downsample = None
+        if stride != 1 or self.in_ch != out_ch:
+            downsample = nn.Sequential(
+                nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
+                norm_layer(out_ch))
+        layers = [Bottleneck(self.in_ch, width, stride, downsample,
+                             norm_layer)]
+        self.in_ch = out_ch
+        for _ in range(1, blocks):
+            layers.append(Bottleneck(out_ch, width, norm_layer=norm_layer))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = self.avgpool(x).flatten(1)
+        return self.fc(x)
+
+
+def resnet50(**kwargs):
+    return ResNet50(**kwargs)

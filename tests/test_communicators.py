@@ -1,0 +1,212 @@
+"""Communicator suite — one parameterized run per communicator class.
+
+Mirrors the reference's biggest test file
+(tests/chainermn_tests/communicator_tests/test_communicator.py): an
+ExampleModel gets known grads, multi_node_mean_grad must produce the exact
+mean; plus bcast_data, send/recv (incl. tuples), every collective, the
+*_obj ops and split.
+"""
+
+import pytest
+import torch
+import torch.nn as nn
+
+import chainermn_amd
+from chainermn_amd.testing import run_distributed
+
+COMM_NAMES = ['naive', 'flat', 'hierarchical', 'two_dimensional',
+              'non_cuda_aware', 'single_node']
+
+
+class ExampleModel(nn.Module):
+
+    def __init__(self):
+        super().__init__()
+        self.a = nn.Linear(2, 3)
+        self.b = nn.Linear(3, 4)
+        self.c = nn.Linear(4, 5)
+
+
+def _set_grads(model, value):
+    for p in model.parameters():
+        p.grad = torch.full_like(p.data, value)
+
+
+def _mean_grad_worker(rank, size, name):
+    comm = chainermn_amd.create_communicator(name)
+    assert comm.rank == rank
+    assert comm.size == size
+    model = ExampleModel()
+    _set_grads(model, float(rank + 1))
+    comm.multi_node_mean_grad(model)
+    expect = sum(range(1, size + 1)) / size
+    for p in model.parameters():
+        assert torch.allclose(p.grad, torch.full_like(p.grad, expect)), \
+            '{}: {} != {}'.format(name, p.grad.flatten()[0], expect)
+    # zero_fill: missing grads are treated as zeros
+    model2 = ExampleModel()
+    if rank == 0:
+        _set_grads(model2, 2.0)
+        model2.c.weight.grad = None
+        model2.c.bias.grad = None
+    else:
+        _set_grads(model2, 0.0)
+    comm.multi_node_mean_grad(model2, zero_fill=True)
+    expect2 = 2.0 / size if rank is not None else 0
+    assert torch.allclose(model2.a.weight.grad,
+                          torch.full_like(model2.a.weight.grad, expect2))
+    assert torch.allclose(model2.c.weight.grad,
+                          torch.zeros_like(model2.c.weight.grad))
+    return True
+
+
+@pytest.mark.parametrize('name', COMM_NAMES)
+def test_mean_grad(name):
+    assert run_distributed(_mean_grad_worker, 2, args=(name,)) == [True, True]
+
+
+def _bcast_data_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    model = ExampleModel()
+    with torch.no_grad():
+        for p in model.parameters():
+            p.data.fill_(float(rank + 7))
+    comm.bcast_data(model)
+    for p in model.parameters():
+        assert torch.allclose(p.data, torch.full_like(p.data, 7.0))
+    return True
+
+
+def test_bcast_data():
+    run_distributed(_bcast_data_worker, 2)
+
+
+def _p2p_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    # single tensor
+    if rank == 0:
+        comm.send(torch.arange(6, dtype=torch.float32).reshape(2, 3), 1)
+    else:
+        x = comm.recv(0)
+        assert x.shape == (2, 3)
+        assert torch.equal(x, torch.arange(6, dtype=torch.float32)
+                           .reshape(2, 3))
+    # tuple payload, reverse direction
+    if rank == 1:
+        comm.send((torch.ones(3), torch.zeros(2, 2, dtype=torch.float64)), 0)
+    else:
+        t = comm.recv(1)
+        assert isinstance(t, tuple) and len(t) == 2
+        assert t[1].dtype == torch.float64
+    return True
+
+
+def test_send_recv():
+    run_distributed(_p2p_worker, 2)
+
+
+def _collectives_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    # allreduce
+    out = comm.allreduce(torch.full((3,), float(rank + 1)))
+    assert torch.allclose(out, torch.full((3,), float(sum(range(1, size + 1)))))
+    # bcast
+    x = torch.arange(4, dtype=torch.float32) if rank == 0 else None
+    out = comm.bcast(x, root=0)
+    assert torch.equal(out, torch.arange(4, dtype=torch.float32))
+    # allgather (different shapes per rank)
+    xs = comm.allgather(torch.full((rank + 1,), float(rank)))
+    assert len(xs) == size
+    for r, t in enumerate(xs):
+        assert t.shape == (r + 1,)
+        assert torch.allclose(t, torch.full((r + 1,), float(r)))
+    # alltoall
+    outs = comm.alltoall([torch.full((2,), float(rank * 10 + j))
+                          for j in range(size)])
+    for src, t in enumerate(outs):
+        assert torch.allclose(t, torch.full((2,), float(src * 10 + rank)))
+    # gather
+    g = comm.gather(torch.full((2,), float(rank)), root=1)
+    if rank == 1:
+        for r, t in enumerate(g):
+            assert torch.allclose(t, torch.full((2,), float(r)))
+    else:
+        assert g is None
+    # scatter
+    if rank == 0:
+        mine = comm.scatter([torch.full((2,), float(10 + r))
+                             for r in range(size)], root=0)
+    else:
+        mine = comm.scatter(None, root=0)
+    assert torch.allclose(mine, torch.full((2,), float(10 + rank)))
+    return True
+
+
+def test_collectives():
+    run_distributed(_collectives_worker, 2)
+
+
+def _obj_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    if rank == 0:
+        comm.send_obj({'hello': [1, 2, 3]}, 1)
+    else:
+        assert comm.recv_obj(0) == {'hello': [1, 2, 3]}
+    assert comm.bcast_obj('payload' if rank == 0 else None) == 'payload'
+    g = comm.gather_obj(rank * 2, root=0)
+    if rank == 0:
+        assert g == [0, 2]
+    assert comm.allgather_obj(rank) == list(range(size))
+    assert comm.allreduce_obj(rank + 1) == sum(range(1, size + 1))
+    assert comm.allreduce_obj({'v': float(rank)}) == \
+        {'v': float(sum(range(size)))}
+    return True
+
+
+def test_obj_comms():
+    run_distributed(_obj_worker, 2)
+
+
+def _split_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    sub = comm.split(color=rank % 2, key=rank)
+    assert sub.size == size // 2 + (1 if rank % 2 < size % 2 else 0)
+    out = sub.allreduce(torch.tensor([float(rank)]))
+    same_color = [r for r in range(size) if r % 2 == rank % 2]
+    assert out.item() == float(sum(same_color))
+    return True
+
+
+def test_split():
+    run_distributed(_split_worker, 4)
+
+
+def _geometry_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    # all processes run on one host in this harness
+    assert comm.intra_rank == rank
+    assert comm.intra_size == size
+    assert comm.inter_rank == 0
+    assert comm.inter_size == 1
+    return True
+
+
+def test_geometry():
+    run_distributed(_geometry_worker, 2)
+
+
+def test_dummy_communicator():
+    comm = chainermn_amd.create_communicator('dummy')
+    assert comm.size == 1 and comm.rank == 0
+    model = ExampleModel()
+    _set_grads(model, 3.0)
+    comm.multi_node_mean_grad(model)
+    for p in model.parameters():
+        assert torch.allclose(p.grad, torch.full_like(p.grad, 3.0))
+    assert comm.allreduce_obj(5) == 5
+    assert comm.bcast(torch.ones(2)) is not None
+
+
+def test_unknown_name():
+    with pytest.raises(ValueError):
+        chainermn_amd.create_communicator('nonexistent')

@@ -1,0 +1,92 @@
+"""Dataset scatter / iterators (reference: datasets_tests +
+iterators_tests): union of shards covers the dataset, balance, equal
+length; multi-node iterator replays identical batches."""
+
+import torch
+
+import chainermn_amd
+from chainermn_amd.datasets import (create_empty_dataset, scatter_dataset,
+                                    scatter_index)
+from chainermn_amd.iterators import (create_multi_node_iterator,
+                                     create_synchronized_iterator)
+from chainermn_amd.testing import run_distributed
+
+
+def _scatter_worker(rank, size, n, shuffle, force_equal):
+    comm = chainermn_amd.create_communicator('naive')
+    dataset = list(range(n))
+    shard = scatter_dataset(dataset, comm, shuffle=shuffle, seed=42,
+                            force_equal_length=force_equal)
+    return list(shard)
+
+
+def test_scatter_dataset_covers_and_balances():
+    for n in (10, 11):
+        shards = run_distributed(_scatter_worker, 2, args=(n, False, False))
+        union = sorted(shards[0] + shards[1])
+        assert union == list(range(n))
+        assert abs(len(shards[0]) - len(shards[1])) <= 1
+
+
+def test_scatter_dataset_equal_length_pads():
+    shards = run_distributed(_scatter_worker, 2, args=(11, False, True))
+    assert len(shards[0]) == len(shards[1]) == 6
+    covered = set(shards[0]) | set(shards[1])
+    assert covered == set(range(11))
+
+
+def test_scatter_dataset_shuffle_seeded():
+    a = run_distributed(_scatter_worker, 2, args=(10, True, False))
+    b = run_distributed(_scatter_worker, 2, args=(10, True, False))
+    assert a == b  # same seed -> same permutation
+    assert sorted(a[0] + a[1]) == list(range(10))
+
+
+def _scatter_index_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    idx = scatter_index(10, comm)
+    return idx
+
+
+def test_scatter_index():
+    shards = run_distributed(_scatter_index_worker, 2)
+    assert sorted(shards[0] + shards[1]) == list(range(10))
+
+
+def test_create_empty_dataset():
+    ds = create_empty_dataset([1, 2, 3])
+    assert len(ds) == 3
+    assert ds[0] == ()
+    try:
+        ds[3]
+        assert False
+    except IndexError:
+        pass
+
+
+def _mn_iterator_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    base = [torch.tensor([i, i + 1]) for i in range(4)] if rank == 0 else None
+    it = create_multi_node_iterator(base, comm, rank_master=0)
+    seen = [b.tolist() for b in it]
+    return seen
+
+
+def test_multi_node_iterator_replays():
+    results = run_distributed(_mn_iterator_worker, 2)
+    assert results[0] == results[1]
+    assert results[0] == [[0, 1], [1, 2], [2, 3], [3, 4]]
+
+
+def _sync_iterator_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    torch.manual_seed(100 + rank)  # deliberately different
+    it = create_synchronized_iterator(range(3), comm)
+    draws = torch.randperm(8).tolist()
+    return (list(it), draws)
+
+
+def test_synchronized_iterator_syncs_rng():
+    results = run_distributed(_sync_iterator_worker, 2)
+    assert results[0][0] == results[1][0] == [0, 1, 2]
+    assert results[0][1] == results[1][1]
