@@ -1,0 +1,55 @@
+// pybind11 bindings for chainermn_amd._C (gfx950 kernels).
+
+#include <torch/extension.h>
+
+void batched_pack(torch::Tensor ptrs, torch::Tensor sizes,
+                  torch::Tensor offsets, torch::Tensor out,
+                  torch::ScalarType src_dtype);
+void batched_unpack(torch::Tensor ptrs, torch::Tensor sizes,
+                    torch::Tensor offsets, torch::Tensor in, double scale,
+                    torch::ScalarType dst_dtype);
+void cast_flat(torch::Tensor src, torch::Tensor dst, double scale);
+void fused_sgd(torch::Tensor p, torch::Tensor p_lowp, torch::Tensor g,
+               torch::Tensor v, double lr, double momentum, double wd,
+               double dampening, bool nesterov, double gscale);
+void fused_adam(torch::Tensor p, torch::Tensor p_lowp, torch::Tensor g,
+                torch::Tensor m, torch::Tensor v, double lr, double b1,
+                double b2, double eps, double wd, long step, bool adamw,
+                double gscale);
+void segment_sq_norms(torch::Tensor flat, torch::Tensor offsets,
+                      torch::Tensor sizes, torch::Tensor out);
+void lars_update(torch::Tensor p, torch::Tensor p_lowp, torch::Tensor g,
+                 torch::Tensor v, torch::Tensor offsets, torch::Tensor sizes,
+                 torch::Tensor p_sqn, torch::Tensor g_sqn, double lr,
+                 double momentum, double wd, double eta, double eps,
+                 double gscale);
+std::tuple<torch::Tensor, torch::Tensor, long> bn_stats(torch::Tensor x);
+std::tuple<torch::Tensor, torch::Tensor> bn_bwd_reduce(torch::Tensor x,
+                                                       torch::Tensor gy,
+                                                       torch::Tensor mean,
+                                                       torch::Tensor rstd);
+torch::Tensor bn_fwd(torch::Tensor x, torch::Tensor mean, torch::Tensor rstd,
+                     torch::Tensor gamma, torch::Tensor beta);
+torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
+                           torch::Tensor mean, torch::Tensor rstd,
+                           torch::Tensor gamma, torch::Tensor gbeta,
+                           torch::Tensor ggamma, double total_count);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "chainermn_amd MI355X (gfx950) kernels";
+  m.def("batched_pack", &batched_pack,
+        "gather param grads into flat wire buffer (fused cast, K1)");
+  m.def("batched_unpack", &batched_unpack,
+        "scatter flat buffer to grads (fused scale+cast, K2)");
+  m.def("cast_flat", &cast_flat, "flat cast with fused scale (K3)");
+  m.def("fused_sgd", &fused_sgd, "flat-arena SGD/momentum step (K5)");
+  m.def("fused_adam", &fused_adam, "flat-arena Adam/AdamW step (K5)");
+  m.def("segment_sq_norms", &segment_sq_norms,
+        "per-layer sum of squares (LARS)");
+  m.def("lars_update", &lars_update, "per-layer trust-ratio update (LARS)");
+  m.def("bn_stats", &bn_stats, "per-channel (sum, sumsq, count) (K6)");
+  m.def("bn_fwd", &bn_fwd, "BN normalize with global stats (K7)");
+  m.def("bn_bwd_reduce", &bn_bwd_reduce,
+        "per-channel (sum gy, sum gy*xhat) (K6)");
+  m.def("bn_bwd_input", &bn_bwd_input, "BN input gradient (K7)");
+}

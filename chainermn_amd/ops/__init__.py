@@ -86,7 +86,7 @@ def pack_grads(plan, out_flat):
     """Gather every param.grad into ``out_flat`` (casting to its dtype)."""
     if out_flat.is_cuda:
         require().batched_pack(plan.grad_ptrs(), plan.sizes, plan.offsets,
-                               out_flat)
+                               out_flat, plan.params[0].grad.dtype)
         return
     out_flat.zero_()
     for p, off, n in zip(plan.params, plan.offsets_cpu, plan.sizes_cpu):
@@ -97,7 +97,8 @@ def unpack_grads(plan, in_flat, scale=1.0):
     """Scatter ``in_flat`` back into param.grads with a fused ``×scale``."""
     if in_flat.is_cuda:
         require().batched_unpack(plan.grad_ptrs(), plan.sizes, plan.offsets,
-                                 in_flat, float(scale))
+                                 in_flat, float(scale),
+                                 plan.params[0].grad.dtype)
         return
     for p, off, n in zip(plan.params, plan.offsets_cpu, plan.sizes_cpu):
         g = in_flat[off:off + n].to(p.grad.dtype) * scale
@@ -116,7 +117,7 @@ def cast_flat(src, dst, scale=1.0):
 
 
 def fused_sgd(p, g, v, lr, momentum=0.0, weight_decay=0.0, dampening=0.0,
-              nesterov=False, grad_scale=1.0):
+              nesterov=False, grad_scale=1.0, p_lowp=None):
     """Flat-buffer SGD/MomentumSGD step (torch.optim.SGD semantics).
 
     ``g`` may be fp32/bf16/fp16 (the all-reduced buffer); ``grad_scale``
@@ -124,7 +125,9 @@ def fused_sgd(p, g, v, lr, momentum=0.0, weight_decay=0.0, dampening=0.0,
     momentum buffer (ignored when momentum == 0).
     """
     if p.is_cuda:
-        require().fused_sgd(p, g, v if v is not None else p.new_empty(0),
+        require().fused_sgd(p, p_lowp if p_lowp is not None
+                            else p.new_empty(0),
+                            g, v if v is not None else p.new_empty(0),
                             float(lr), float(momentum), float(weight_decay),
                             float(dampening), bool(nesterov),
                             float(grad_scale))
@@ -136,13 +139,17 @@ def fused_sgd(p, g, v, lr, momentum=0.0, weight_decay=0.0, dampening=0.0,
         v.mul_(momentum).add_(grad, alpha=1.0 - dampening)
         grad = grad.add(v, alpha=momentum) if nesterov else v
     p.add_(grad, alpha=-lr)
+    if p_lowp is not None:
+        p_lowp.copy_(p.to(p_lowp.dtype))
 
 
 def fused_adam(p, g, m, v, lr, beta1, beta2, eps, weight_decay=0.0,
-               step=1, adamw=False, grad_scale=1.0):
+               step=1, adamw=False, grad_scale=1.0, p_lowp=None):
     """Flat-buffer Adam/AdamW step (torch.optim.Adam semantics)."""
     if p.is_cuda:
-        require().fused_adam(p, g, m, v, float(lr), float(beta1),
+        require().fused_adam(p, p_lowp if p_lowp is not None
+                             else p.new_empty(0),
+                             g, m, v, float(lr), float(beta1),
                              float(beta2), float(eps), float(weight_decay),
                              int(step), bool(adamw), float(grad_scale))
         return
@@ -157,6 +164,8 @@ def fused_adam(p, g, m, v, lr, beta1, beta2, eps, weight_decay=0.0,
     bc2 = 1.0 - beta2 ** step
     denom = (v / bc2).sqrt_().add_(eps)
     p.addcdiv_(m, denom, value=-lr / bc1)
+    if p_lowp is not None:
+        p_lowp.copy_(p.to(p_lowp.dtype))
 
 
 def segment_sq_norms(flat, offsets, sizes, out):
@@ -171,14 +180,16 @@ def segment_sq_norms(flat, offsets, sizes, out):
 
 
 def lars_update(p, g, v, offsets, sizes, p_sqn, g_sqn, lr, momentum,
-                weight_decay, eta, eps, grad_scale=1.0):
+                weight_decay, eta, eps, grad_scale=1.0, p_lowp=None):
     """Per-layer trust-ratio (LARS) momentum update over the flat arenas.
 
     trust = eta * ||p|| / (||g|| + weight_decay * ||p|| + eps) per segment;
     v = momentum * v + trust * lr * (g + wd * p); p -= v.
     """
     if p.is_cuda:
-        require().lars_update(p, g, v, offsets, sizes, p_sqn, g_sqn,
+        require().lars_update(p, p_lowp if p_lowp is not None
+                              else p.new_empty(0),
+                              g, v, offsets, sizes, p_sqn, g_sqn,
                               float(lr), float(momentum), float(weight_decay),
                               float(eta), float(eps), float(grad_scale))
         return
@@ -197,3 +208,5 @@ def lars_update(p, g, v, offsets, sizes, p_sqn, g_sqn, lr, momentum,
         upd = gs.add(ps, alpha=weight_decay)
         vs.mul_(momentum).add_(upd, alpha=trust * lr)
         ps.sub_(vs)
+    if p_lowp is not None:
+        p_lowp.copy_(p.to(p_lowp.dtype))

@@ -1,0 +1,229 @@
+"""Fused data-parallel optimizers — the MI355X flagship gradient path.
+
+This is the structural redesign of the reference's hot loop (SURVEY.md
+§3.2/§3.3).  The reference per step: extract grads → batched pack kernel →
+cast fp16 → ncclAllReduce → fused cast-back+scale → unpack → per-param
+optimizer kernels (≈161 launches).  Here, by construction:
+
+* every parameter lives in one flat low-precision arena (bf16) and every
+  ``param.grad`` is a view into one flat bf16 grad arena, so backward
+  ACCUMULATES the wire buffer directly — pack and cast cost zero;
+* the all-reduce is RCCL over xGMI, in place on the grad arena, on a side
+  HIP stream, double-buffered (grads of step n reduce while step n+1
+  computes — the reference's _DoubleBufferingOptimizer semantics, 1-step
+  stale, without its worker thread);
+* the whole-model update is ONE fused CDNA4 kernel (optim.hip): fp32
+  master params + momentum + mean-scale + bf16 mirror write in a single
+  HBM pass.
+
+Per step: 1 RCCL all-reduce (overlapped) + 1 fused kernel.  The wire
+volume for ResNet-50 is 51 MB bf16 (reference: 102 MB fp32 or 51 MB fp16
+plus two extra full pack/unpack HBM round-trips).
+"""
+
+import torch
+import torch.distributed as dist
+
+from chainermn_amd import ops
+
+
+class _FlatState(object):
+    """Flattens a model's params into arenas and rebinds p.data/p.grad."""
+
+    def __init__(self, model, comm, grad_dtype=torch.bfloat16,
+                 n_grad_buffers=2):
+        params = [p for _, p in sorted(model.named_parameters())
+                  if p.requires_grad]
+        if not params:
+            raise ValueError('model has no trainable parameters')
+        dev = params[0].device
+        dtype = params[0].dtype
+        for p in params:
+            if p.dtype != dtype:
+                raise ValueError('FusedDataParallel requires uniform '
+                                 'parameter dtype')
+        self.comm = comm
+        self.device = dev
+        self.model_dtype = dtype
+        # autograd accumulates into p.grad views, so the grad (wire) arena
+        # must match the param dtype — bf16 params give bf16 wire traffic
+        # for free; fp32 models reduce fp32 (compressed-wire fp32 models go
+        # through PureRcclCommunicator's pack+cast path instead).
+        if grad_dtype != dtype:
+            grad_dtype = dtype
+        self.grad_dtype = grad_dtype
+        self.plan = ops.PackPlan(params, dev)
+        n = self.plan.total
+        # fp32 master + low-precision mirror (or fp32-only when the model
+        # itself is fp32)
+        self.master = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.lowp = None
+        if dtype != torch.float32:
+            self.lowp = torch.zeros(n, dtype=dtype, device=dev)
+        self.grads = [torch.zeros(n, dtype=grad_dtype, device=dev)
+                      for _ in range(n_grad_buffers)]
+        self.params = params
+        # move params into the arena
+        with torch.no_grad():
+            for p, off, sz in zip(params, self.plan.offsets_cpu,
+                                  self.plan.sizes_cpu):
+                self.master[off:off + sz].copy_(p.data.reshape(-1).float())
+                arena = self.lowp if self.lowp is not None else self.master
+                view = arena[off:off + sz].view_as(p.data)
+                view.copy_(p.data)
+                p.data = view
+        self.cur = 0
+        self._bind_grads(0)
+
+    def _bind_grads(self, idx):
+        for p, off, sz in zip(self.params, self.plan.offsets_cpu,
+                              self.plan.sizes_cpu):
+            p.grad = self.grads[idx][off:off + sz].view_as(p.data)
+
+    def grad_arena(self, idx=None):
+        return self.grads[self.cur if idx is None else idx]
+
+    def sync_master_from_params(self):
+        """Refresh fp32 master after an external param mutation
+        (e.g. bcast_data)."""
+        with torch.no_grad():
+            for p, off, sz in zip(self.params, self.plan.offsets_cpu,
+                                  self.plan.sizes_cpu):
+                self.master[off:off + sz].copy_(p.data.reshape(-1).float())
+
+
+class FusedDataParallelSGD(object):
+    """SGD/MomentumSGD over flat arenas with overlapped bf16 all-reduce.
+
+    ``overlap=True`` → double-buffered (1-step-stale) gradients: the
+    all-reduce of step n runs on a side stream under the fwd/bwd of step
+    n+1; the first step applies no update.  ``overlap=False`` → reduce and
+    update synchronously (0 staleness).
+    """
+
+    def __init__(self, model, comm, lr, momentum=0.9, weight_decay=0.0,
+                 dampening=0.0, nesterov=False, overlap=True,
+                 grad_dtype=torch.bfloat16):
+        self.comm = comm
+        self.lr = lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self.dampening = dampening
+        self.nesterov = nesterov
+        self.overlap = overlap
+        self.state = _FlatState(model, comm,
+                                grad_dtype=grad_dtype,
+                                n_grad_buffers=2 if overlap else 1)
+        n = self.state.plan.total
+        self.v = torch.zeros(n, dtype=torch.float32,
+                             device=self.state.device) \
+            if momentum else None
+        if overlap:
+            self.stream = torch.cuda.Stream()
+            self.ev_ar = [torch.cuda.Event(), torch.cuda.Event()]
+        self._step = 0
+
+    def zero_grad(self, set_to_none=False):
+        # current arena is zeroed inside step(); only needed before the
+        # very first backward or after manual grad fiddling
+        if self._step == 0:
+            self.state.grad_arena().zero_()
+
+    def _reduce_async(self, idx):
+        g = self.state.grads[idx]
+        main = torch.cuda.current_stream()
+        self.stream.wait_stream(main)
+        with torch.cuda.stream(self.stream):
+            dist.all_reduce(g, group=self.comm.group)
+            self.ev_ar[idx].record(self.stream)
+
+    def _apply(self, g):
+        ops.fused_sgd(self.state.master, g, self.v, self.lr,
+                      momentum=self.momentum,
+                      weight_decay=self.weight_decay,
+                      dampening=self.dampening, nesterov=self.nesterov,
+                      grad_scale=1.0 / max(self.comm.size, 1),
+                      p_lowp=self.state.lowp)
+
+    def step(self):
+        st = self.state
+        if not self.overlap:
+            g = st.grad_arena()
+            if self.comm.size > 1:
+                dist.all_reduce(g, group=self.comm.group)
+            self._apply(g)
+            g.zero_()
+            self._step += 1
+            return
+        cur, prev = st.cur, 1 - st.cur
+        main = torch.cuda.current_stream()
+        if self.comm.size > 1:
+            self._reduce_async(cur)
+        else:
+            self.ev_ar[cur].record(main)
+        if self._step > 0:
+            main.wait_event(self.ev_ar[prev])
+            self._apply(st.grads[prev])
+            st.grads[prev].zero_()
+        # next backward accumulates into `prev` arena
+        st._bind_grads(prev)
+        st.cur = prev
+        self._step += 1
+
+    def wait(self):
+        if self.overlap and self._step > 0:
+            torch.cuda.current_stream().wait_event(
+                self.ev_ar[1 - self.state.cur])
+
+    # torch.optim-ish compat
+    @property
+    def param_groups(self):
+        return [{'lr': self.lr, 'momentum': self.momentum,
+                 'weight_decay': self.weight_decay,
+                 'params': self.state.params}]
+
+    def state_dict(self):
+        return {'step': self._step, 'master': self.state.master,
+                'v': self.v, 'lr': self.lr}
+
+    def load_state_dict(self, sd):
+        self._step = sd['step']
+        self.state.master.copy_(sd['master'])
+        if self.v is not None and sd['v'] is not None:
+            self.v.copy_(sd['v'])
+
+
+class FusedDataParallelLARS(FusedDataParallelSGD):
+    """Layer-wise adaptive rate scaling (You et al. 2017) over the same
+    arenas — the large-batch config (BASELINE.json config 5).  Per layer:
+    trust = eta·||p||/(||g|| + wd·||p|| + eps); v = m·v + trust·lr·(g+wd·p).
+    """
+
+    def __init__(self, model, comm, lr, momentum=0.9, weight_decay=5e-5,
+                 eta=0.001, eps=1e-9, overlap=True,
+                 grad_dtype=torch.bfloat16):
+        super().__init__(model, comm, lr, momentum=momentum,
+                         weight_decay=weight_decay, overlap=overlap,
+                         grad_dtype=grad_dtype)
+        if self.v is None:
+            n = self.state.plan.total
+            self.v = torch.zeros(n, dtype=torch.float32,
+                                 device=self.state.device)
+        self.eta = eta
+        self.eps = eps
+        nseg = len(self.state.plan.sizes_cpu)
+        self.p_sqn = torch.zeros(nseg, dtype=torch.float32,
+                                 device=self.state.device)
+        self.g_sqn = torch.zeros(nseg, dtype=torch.float32,
+                                 device=self.state.device)
+
+    def _apply(self, g):
+        st = self.state
+        plan = st.plan
+        ops.segment_sq_norms(st.master, plan.offsets, plan.sizes, self.p_sqn)
+        ops.segment_sq_norms(g, plan.offsets, plan.sizes, self.g_sqn)
+        ops.lars_update(st.master, g, self.v, plan.offsets, plan.sizes,
+                        self.p_sqn, self.g_sqn, self.lr, self.momentum,
+                        self.weight_decay, self.eta, self.eps,
+                        grad_scale=1.0 / max(self.comm.size, 1),
+                        p_lowp=st.lowp)
