@@ -1,20 +1,21 @@
-// Multi-node BatchNorm device kernels (gfx950).
+// BatchNorm device kernels (gfx950) — used by MultiNodeBatchNormalization
+// (cross-rank global-batch BN) and by the local
+// FusedBatchNorm module (same kernels, no all-reduce).
 //
-// Rebuild of SURVEY.md §2.4 K6/K7: the reference computed local BN
-// statistics with CuPy reductions and normalized with Chainer/cuDNN
-// kernels.  Here:
 //   bn_stats      : per-channel (Σx, Σx²) in ONE fused pass (the 2×C+1
-//                   buffer that goes over RCCL), fp32 accumulation,
-//                   wave-shuffle + LDS-tree partials, one atomicAdd per
-//                   (block, channel)
-//   bn_fwd        : y = (x-μ)·rstd·γ + β elementwise, reads the
-//                   all-reduced stats
+//                   buffer that goes over RCCL), fp32 accumulation
+//   bn_fwd        : y = (x-μ)·rstd·γ + β elementwise
 //   bn_bwd_reduce : per-channel (Σgy, Σgy·x̂) fused pass
 //   bn_bwd_input  : gx = γ·rstd·(gy − (x̂·Σgy·x̂ + Σgy)/M) elementwise
 //
-// Both NCHW-contiguous and channels-last layouts are native: channels-last
-// (the MIOpen-friendly conv layout) reduces columns of an [M, C] matrix —
-// lanes own channels, accesses stay 16 B-coalesced with zero bank traffic.
+// Perf notes (measured against torch's native channels_last BN kernels,
+// which cost 66% of a ResNet-50 step on this chip — profiles/):
+//  * channels-last reduce: lanes own 4 channels (uint2 = 4×bf16 loads),
+//    per-thread fp32 accumulators over rows, LDS cross-row tree, ≤312
+//    blocks → at most a few hundred atomics per channel
+//  * pointwise: 8 elems/lane (uint4 for 16-bit dtypes, 2×float4 for f32)
+//    on both layouts when C (channels-last) or S (NCHW) is a multiple
+//    of 8 — ResNet shapes always are; scalar fallback otherwise
 
 #include <torch/extension.h>
 #include <ATen/Dispatch.h>
@@ -76,9 +77,28 @@ __global__ void bn_reduce_nchw_kernel(const T* __restrict__ x,
   }
 }
 
+// 4-channel vector load helpers (channels-last reduce path)
+__device__ __forceinline__ void load4(const float* p, float (&r)[4]) {
+  const float4 v = *reinterpret_cast<const float4*>(p);
+  r[0] = v.x; r[1] = v.y; r[2] = v.z; r[3] = v.w;
+}
+template <typename T16>
+__device__ __forceinline__ void load4(const T16* p, float (&r)[4]) {
+  const uint2 v = *reinterpret_cast<const uint2*>(p);
+  const unsigned u[2] = {v.x, v.y};
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    r[2 * h] = float(T16(uint16_t(u[h] & 0xffffu),
+                         typename T16::from_bits_t()));
+    r[2 * h + 1] = float(T16(uint16_t(u[h] >> 16),
+                             typename T16::from_bits_t()));
+  }
+}
+
 // ---------- channels-last: x[m][c] (m = N*S), reduce over m --------------
-// Thread t owns channels [4*(t%ct), 4*(t%ct)+4) within a channel tile;
-// rows are covered by (blockIdx, t/ct).  C % 4 == 0 path.
+// CT threads per row (each owning 4 consecutive channels), RPB = 256/CT
+// rows in flight per block; per-thread fp32 accumulators; LDS tree over
+// the RPB row-threads; one atomicAdd per (block, channel).
 template <typename T, bool BWD>
 __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
                                     const T* __restrict__ gy,
@@ -87,120 +107,105 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
                                     float* __restrict__ out0,
                                     float* __restrict__ out1,
                                     int C, long M) {
-  // channel tile of 1024 (=256 threads × 4); loop tiles when C > 1024.
-  const int ct = min(C / 4, (int)blockDim.x);          // threads per row
-  const int rows_per_block = blockDim.x / ct;
+  __shared__ float lds[CMN_BLOCK * 8];  // 8 KiB: [tid][8]
+  const int ct = min(C / 4, (int)blockDim.x);
+  const int rpb = blockDim.x / ct;
   const int tile_ch = ct * 4;
+  const int tc = threadIdx.x % ct;
+  const int tr = threadIdx.x / ct;
   for (int c_base = 0; c_base < C; c_base += tile_ch) {
-    const int tc = threadIdx.x % ct;
-    const int tr = threadIdx.x / ct;
     const int c = c_base + tc * 4;
-    float a[8] = {0, 0, 0, 0, 0, 0, 0, 0};  // 4×(sum, sq)
-    if (c + 4 <= C && tr < rows_per_block) {
-      float m4[4] = {0, 0, 0, 0}, r4[4] = {0, 0, 0, 0};
-      if (BWD) {
-#pragma unroll
-        for (int k = 0; k < 4; ++k) { m4[k] = mean[c + k]; r4[k] = rstd[c + k]; }
-      }
-      const long row_stride = (long)gridDim.x * rows_per_block;
-      for (long m = blockIdx.x * (long)rows_per_block + tr; m < M;
-           m += row_stride) {
-        const long idx = m * C + c;
-        if (BWD) {
-          const uint2 vg = *reinterpret_cast<const uint2*>(gy + idx);
-          const uint2 vx = *reinterpret_cast<const uint2*>(x + idx);
-          const unsigned ug[2] = {vg.x, vg.y}, ux[2] = {vx.x, vx.y};
-#pragma unroll
-          for (int h = 0; h < 2; ++h) {
-#pragma unroll
-            for (int l = 0; l < 2; ++l) {
-              const int k = 2 * h + l;
-              const T gv = T(uint16_t(l ? (ug[h] >> 16) : (ug[h] & 0xffffu)),
-                             typename T::from_bits_t());
-              const T xv = T(uint16_t(l ? (ux[h] >> 16) : (ux[h] & 0xffffu)),
-                             typename T::from_bits_t());
-              const float g = float(gv);
-              a[2 * k] += g;
-              a[2 * k + 1] += g * ((float(xv) - m4[k]) * r4[k]);
-            }
-          }
-        } else {
-          const uint2 vx = *reinterpret_cast<const uint2*>(x + idx);
-          const unsigned ux[2] = {vx.x, vx.y};
-#pragma unroll
-          for (int h = 0; h < 2; ++h) {
-#pragma unroll
-            for (int l = 0; l < 2; ++l) {
-              const int k = 2 * h + l;
-              const T xv = T(uint16_t(l ? (ux[h] >> 16) : (ux[h] & 0xffffu)),
-                             typename T::from_bits_t());
-              const float v = float(xv);
-              a[2 * k] += v;
-              a[2 * k + 1] += v * v;
-            }
-          }
-        }
-      }
+    float a[8] = {0, 0, 0, 0, 0, 0, 0, 0};  // 4×(sum0, sum1)
+    float m4[4] = {0, 0, 0, 0}, r4[4] = {0, 0, 0, 0};
+    if (BWD) {
 #pragma unroll
       for (int k = 0; k < 4; ++k) {
-        if (a[2 * k] != 0.f || a[2 * k + 1] != 0.f) {
-          atomicAdd(&out0[c + k], a[2 * k]);
-          atomicAdd(&out1[c + k], a[2 * k + 1]);
+        m4[k] = mean[c + k];
+        r4[k] = rstd[c + k];
+      }
+    }
+    const long row_stride = (long)gridDim.x * rpb;
+    if (tr < rpb) {
+      for (long m = blockIdx.x * (long)rpb + tr; m < M; m += row_stride) {
+        const long idx = m * C + c;
+        float xv[4], gv[4];
+        load4(x + idx, xv);
+        if (BWD) load4(gy + idx, gv);
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          if (BWD) {
+            a[2 * k] += gv[k];
+            a[2 * k + 1] += gv[k] * ((xv[k] - m4[k]) * r4[k]);
+          } else {
+            a[2 * k] += xv[k];
+            a[2 * k + 1] += xv[k] * xv[k];
+          }
         }
       }
     }
-  }
-}
-
-// fp32 channels-last falls back to a scalar-indexed variant.
-template <typename T, bool BWD>
-__global__ void bn_reduce_cl_scalar_kernel(const T* __restrict__ x,
-                                           const T* __restrict__ gy,
-                                           const float* __restrict__ mean,
-                                           const float* __restrict__ rstd,
-                                           float* __restrict__ out0,
-                                           float* __restrict__ out1,
-                                           int C, long M) {
-  const int c = blockIdx.x % C;
-  const int blk = blockIdx.x / C;
-  const int nblk = gridDim.x / C;
-  if (blk >= nblk) return;
-  __shared__ float scratch[2][CMN_BLOCK / CMN_WAVE];
-  float a0 = 0.f, a1 = 0.f;
-  const float mu = BWD ? mean[c] : 0.f;
-  const float rs = BWD ? rstd[c] : 0.f;
-  const long stride = (long)nblk * blockDim.x;
-  for (long m = blk * (long)blockDim.x + threadIdx.x; m < M; m += stride) {
-    const long idx = m * C + c;
-    if (BWD) {
-      const float g = float(gy[idx]);
-      a0 += g;
-      a1 += g * ((float(x[idx]) - mu) * rs);
-    } else {
-      const float v = float(x[idx]);
-      a0 += v;
-      a1 += v * v;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = a[k];
+    __syncthreads();
+    // cross-row tree: threads with tr < off accumulate row tr+off
+    for (int off = rpb >> 1; off > 0; off >>= 1) {
+      if (tr < off) {
+#pragma unroll
+        for (int k = 0; k < 8; ++k)
+          lds[threadIdx.x * 8 + k] += lds[(threadIdx.x + off * ct) * 8 + k];
+      }
+      __syncthreads();
     }
-  }
-  const int lane = threadIdx.x & (CMN_WAVE - 1);
-  const int wid = threadIdx.x / CMN_WAVE;
-  a0 = wave_reduce_sum(a0);
-  a1 = wave_reduce_sum(a1);
-  if (lane == 0) { scratch[0][wid] = a0; scratch[1][wid] = a1; }
-  __syncthreads();
-  if (wid == 0) {
-    const int nw = blockDim.x / CMN_WAVE;
-    float r0 = (lane < nw) ? scratch[0][lane] : 0.f;
-    float r1 = (lane < nw) ? scratch[1][lane] : 0.f;
-    r0 = wave_reduce_sum(r0);
-    r1 = wave_reduce_sum(r1);
-    if (lane == 0) { atomicAdd(&out0[c], r0); atomicAdd(&out1[c], r1); }
+    if (tr == 0) {
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        atomicAdd(&out0[c + k], lds[threadIdx.x * 8 + 2 * k]);
+        atomicAdd(&out1[c + k], lds[threadIdx.x * 8 + 2 * k + 1]);
+      }
+    }
+    __syncthreads();
   }
 }
 
 // ---------- elementwise normalize / input-grad ---------------------------
 // MODE 0: y = (x-μ)·rstd·γ + β
 // MODE 1: gx = γ·rstd·(gy − (x̂·sgx + sg)/M)
+// Vectorized: 8 elems/lane.  CHANNELS_LAST → the 8 elems are 8 consecutive
+// channels (requires C % 8 == 0); NCHW → same channel (requires S % 8 == 0).
+template <typename T, int MODE, bool CHANNELS_LAST>
+__global__ void bn_pointwise_vec_kernel(const T* __restrict__ x,
+                                        const T* __restrict__ gy,
+                                        T* __restrict__ out,
+                                        const float* __restrict__ mean,
+                                        const float* __restrict__ rstd,
+                                        const float* __restrict__ gamma,
+                                        const float* __restrict__ beta,
+                                        const float* __restrict__ sg,
+                                        const float* __restrict__ sgx,
+                                        float inv_m, int C, long S,
+                                        long total8) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long v = blockIdx.x * (long)blockDim.x + threadIdx.x; v < total8;
+       v += stride) {
+    const long base = v * 8;
+    float xv[8], gv[8], o[8];
+    V8<T>::load(x + base, xv);
+    if (MODE == 1) V8<T>::load(gy + base, gv);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      const int c = CHANNELS_LAST ? (int)((base + k) % C)
+                                  : (int)((base / S) % C);
+      const float xh = (xv[k] - mean[c]) * rstd[c];
+      if (MODE == 0) {
+        o[k] = xh * gamma[c] + beta[c];
+      } else {
+        o[k] = gamma[c] * rstd[c]
+               * (gv[k] - (xh * sgx[c] + sg[c]) * inv_m);
+      }
+    }
+    V8<T>::store(out + base, o);
+  }
+}
+
 template <typename T, int MODE, bool CHANNELS_LAST>
 __global__ void bn_pointwise_kernel(const T* __restrict__ x,
                                     const T* __restrict__ gy,
@@ -209,19 +214,18 @@ __global__ void bn_pointwise_kernel(const T* __restrict__ x,
                                     const float* __restrict__ rstd,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ beta,
-                                    const float* __restrict__ sg,   // Σgy
-                                    const float* __restrict__ sgx,  // Σgy·x̂
+                                    const float* __restrict__ sg,
+                                    const float* __restrict__ sgx,
                                     float inv_m, int C, long S, long total) {
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
        i += stride) {
     const int c = CHANNELS_LAST ? (int)(i % C) : (int)((i / S) % C);
     const float xv = float(x[i]);
+    const float xh = (xv - mean[c]) * rstd[c];
     if (MODE == 0) {
-      const float xh = (xv - mean[c]) * rstd[c];
       out[i] = T(xh * gamma[c] + beta[c]);
     } else {
-      const float xh = (xv - mean[c]) * rstd[c];
       const float g = float(gy[i]);
       out[i] = T(gamma[c] * rstd[c] * (g - (xh * sgx[c] + sg[c]) * inv_m));
     }
@@ -250,7 +254,7 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
   auto stream = c10::hip::getCurrentHIPStream();
   out0.zero_();
   out1.zero_();
-  const bool cl = is_channels_last(x);
+  const bool cl = is_channels_last(x) || x.dim() == 2;
   const bool bwd = gy != nullptr;
   DISPATCH_X(x, [&] {
     const scalar_t* xp = (const scalar_t*)x.data_ptr();
@@ -259,30 +263,20 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
     const float* rp = bwd ? rstd->data_ptr<float>() : nullptr;
     auto run = [&](auto bwd_tag) {
       constexpr bool IS_BWD = decltype(bwd_tag)::value;
-      if (cl) {
-        constexpr bool is16 = std::is_same<scalar_t, c10::BFloat16>::value ||
-                              std::is_same<scalar_t, c10::Half>::value;
-        if (is16 && C % 4 == 0) {
-          if constexpr (is16) {
-            const int ct = std::min<long>(C / 4, CMN_BLOCK);
-            const int rpb = CMN_BLOCK / ct;
-            const int grid = (int)std::min<long>((M + rpb - 1) / rpb, 1024);
-            hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD>),
-                               dim3(grid), dim3(CMN_BLOCK), 0,
-                               stream.stream(), xp, gp, mp, rp,
-                               out0.data_ptr<float>(),
-                               out1.data_ptr<float>(), C, M);
-          }
-        } else {
-          const int nblk = (int)std::min<long>(
-              std::max<long>(2048 / C, 1L),
-              (M + CMN_BLOCK - 1) / CMN_BLOCK);
-          hipLaunchKernelGGL((bn_reduce_cl_scalar_kernel<scalar_t, IS_BWD>),
-                             dim3(C * nblk), dim3(CMN_BLOCK), 0,
-                             stream.stream(), xp, gp, mp, rp,
-                             out0.data_ptr<float>(), out1.data_ptr<float>(),
-                             C, M);
-        }
+      if (cl && C % 4 == 0) {
+        const int ct = std::min<long>(C / 4, CMN_BLOCK);
+        const int rpb = CMN_BLOCK / ct;
+        const int grid = (int)std::min<long>((M + rpb - 1) / rpb, 312);
+        hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD>),
+                           dim3(grid), dim3(CMN_BLOCK), 0, stream.stream(),
+                           xp, gp, mp, rp, out0.data_ptr<float>(),
+                           out1.data_ptr<float>(), C, M);
+      } else if (cl) {
+        // C not divisible by 4: per-channel scalar blocks (cold path)
+        hipLaunchKernelGGL((bn_reduce_nchw_kernel<scalar_t, IS_BWD>),
+                           dim3(C, 8), dim3(CMN_BLOCK), 0, stream.stream(),
+                           xp, gp, mp, rp, out0.data_ptr<float>(),
+                           out1.data_ptr<float>(), C, M, 1L);
       } else {
         const long N = x.size(0);
         const long S = M / N;
@@ -324,28 +318,62 @@ std::tuple<torch::Tensor, torch::Tensor> bn_bwd_reduce(torch::Tensor x,
   return {gbeta, ggamma};
 }
 
+static void bn_pointwise(int mode, const torch::Tensor& x,
+                         const torch::Tensor* gy, torch::Tensor& out,
+                         const torch::Tensor& mean, const torch::Tensor& rstd,
+                         const torch::Tensor& gamma,
+                         const torch::Tensor* beta, const torch::Tensor* sg,
+                         const torch::Tensor* sgx, float inv_m) {
+  const bool cl = is_channels_last(x) || x.dim() == 2;
+  const int C = (int)x.size(1);
+  const long total = x.numel();
+  const long S = (x.dim() > 2 && !cl) ? total / (x.size(0) * C) : 1;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const bool vec = cl ? (C % 8 == 0) : (S % 8 == 0);
+  DISPATCH_X(x, [&] {
+    const scalar_t* xp = (const scalar_t*)x.data_ptr();
+    const scalar_t* gp = gy ? (const scalar_t*)gy->data_ptr() : nullptr;
+    scalar_t* op = (scalar_t*)out.data_ptr();
+    const float* bp = beta ? beta->data_ptr<float>() : nullptr;
+    const float* sgp = sg ? sg->data_ptr<float>() : nullptr;
+    const float* sgxp = sgx ? sgx->data_ptr<float>() : nullptr;
+    auto launch = [&](auto mode_tag, auto cl_tag) {
+      constexpr int MODE = decltype(mode_tag)::value;
+      constexpr bool CL = decltype(cl_tag)::value;
+      if (vec) {
+        hipLaunchKernelGGL((bn_pointwise_vec_kernel<scalar_t, MODE, CL>),
+                           dim3(cmn_grid(total / 8, CMN_BLOCK)),
+                           dim3(CMN_BLOCK), 0, stream.stream(), xp, gp, op,
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), bp, sgp, sgxp, inv_m, C,
+                           S, total / 8);
+      } else {
+        hipLaunchKernelGGL((bn_pointwise_kernel<scalar_t, MODE, CL>),
+                           dim3(cmn_grid(total, CMN_BLOCK)), dim3(CMN_BLOCK),
+                           0, stream.stream(), xp, gp, op,
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), bp, sgp, sgxp, inv_m, C,
+                           S, total);
+      }
+    };
+    using i0 = std::integral_constant<int, 0>;
+    using i1 = std::integral_constant<int, 1>;
+    if (mode == 0) {
+      if (cl) launch(i0{}, std::true_type{});
+      else launch(i0{}, std::false_type{});
+    } else {
+      if (cl) launch(i1{}, std::true_type{});
+      else launch(i1{}, std::false_type{});
+    }
+  });
+}
+
 torch::Tensor bn_fwd(torch::Tensor x, torch::Tensor mean, torch::Tensor rstd,
                      torch::Tensor gamma, torch::Tensor beta) {
   TORCH_CHECK(x.is_cuda());
-  const bool cl = is_channels_last(x);
   auto y = torch::empty_like(x);
-  const int C = (int)x.size(1);
-  const long total = x.numel();
-  const long S = (x.dim() > 2 && !cl) ? x.numel() / (x.size(0) * C) : 1;
-  auto stream = c10::hip::getCurrentHIPStream();
-  DISPATCH_X(x, [&] {
-    auto launch = [&](auto cl_tag) {
-      hipLaunchKernelGGL(
-          (bn_pointwise_kernel<scalar_t, 0, decltype(cl_tag)::value>),
-          dim3(cmn_grid(total, CMN_BLOCK)), dim3(CMN_BLOCK), 0,
-          stream.stream(), (const scalar_t*)x.data_ptr(), nullptr,
-          (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
-          rstd.data_ptr<float>(), gamma.data_ptr<float>(),
-          beta.data_ptr<float>(), nullptr, nullptr, 0.f, C, S, total);
-    };
-    if (cl) launch(std::true_type{});
-    else launch(std::false_type{});
-  });
+  bn_pointwise(0, x, nullptr, y, mean, rstd, gamma, &beta, nullptr, nullptr,
+               0.f);
   return y;
 }
 
@@ -354,26 +382,8 @@ torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
                            torch::Tensor gamma, torch::Tensor gbeta,
                            torch::Tensor ggamma, double total_count) {
   TORCH_CHECK(x.is_cuda());
-  const bool cl = is_channels_last(x);
   auto gx = torch::empty_like(x);
-  const int C = (int)x.size(1);
-  const long total = x.numel();
-  const long S = (x.dim() > 2 && !cl) ? x.numel() / (x.size(0) * C) : 1;
-  auto stream = c10::hip::getCurrentHIPStream();
-  DISPATCH_X(x, [&] {
-    auto launch = [&](auto cl_tag) {
-      hipLaunchKernelGGL(
-          (bn_pointwise_kernel<scalar_t, 1, decltype(cl_tag)::value>),
-          dim3(cmn_grid(total, CMN_BLOCK)), dim3(CMN_BLOCK), 0,
-          stream.stream(), (const scalar_t*)x.data_ptr(),
-          (const scalar_t*)gy.data_ptr(), (scalar_t*)gx.data_ptr(),
-          mean.data_ptr<float>(), rstd.data_ptr<float>(),
-          gamma.data_ptr<float>(), nullptr, gbeta.data_ptr<float>(),
-          ggamma.data_ptr<float>(), (float)(1.0 / total_count), C, S,
-          total);
-    };
-    if (cl) launch(std::true_type{});
-    else launch(std::false_type{});
-  });
+  bn_pointwise(1, x, &gy, gx, mean, rstd, gamma, nullptr, &gbeta, &ggamma,
+               (float)(1.0 / total_count));
   return gx;
 }

@@ -337,3 +337,39 @@ def test_bench_smoke():
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     mod.smoke()
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize('cl', [False, True])
+def test_fused_batch_norm_vs_torch(dtype, cl):
+    from chainermn_amd.links import FusedBatchNorm2d
+    torch.manual_seed(10)
+    N, C, H, W = 8, 64, 14, 14
+    x32 = torch.randn(N, C, H, W, device=DEV)
+    bn = FusedBatchNorm2d(C, eps=1e-5, momentum=0.1).to(DEV)
+    ref = torch.nn.BatchNorm2d(C, eps=1e-5, momentum=0.1).to(DEV)
+    x = x32.to(dtype).clone()
+    if cl:
+        x = x.to(memory_format=torch.channels_last)
+    x.requires_grad_(True)
+    xr = x32.clone().requires_grad_(True)
+    y = bn(x)
+    yr = ref(xr)
+    gy = torch.randn_like(yr)
+    y.backward(gy.to(dtype) if not cl
+               else gy.to(dtype).to(memory_format=torch.channels_last))
+    yr.backward(gy)
+    tol = 5e-2 if dtype == torch.bfloat16 else 1e-4
+    assert torch.allclose(y.float(), yr.detach(), atol=tol)
+    assert torch.allclose(x.grad.float(), xr.grad, atol=tol)
+    assert torch.allclose(bn.weight.grad, ref.weight.grad, atol=tol * 20,
+                          rtol=1e-2)
+    assert torch.allclose(bn.bias.grad, ref.bias.grad, atol=tol * 20,
+                          rtol=1e-2)
+    assert torch.allclose(bn.running_mean, ref.running_mean, atol=1e-3)
+    assert torch.allclose(bn.running_var, ref.running_var, atol=1e-3)
+    # eval mode path
+    bn.eval(); ref.eval()
+    ye = bn(x32.to(dtype))
+    yre = ref(x32)
+    assert torch.allclose(ye.float(), yre, atol=tol)
