@@ -24,6 +24,12 @@ void lars_update(torch::Tensor p, torch::Tensor p_lowp, torch::Tensor g,
                  double momentum, double wd, double eta, double eps,
                  double gscale);
 std::tuple<torch::Tensor, torch::Tensor, long> bn_stats(torch::Tensor x);
+void bn_stats_into(torch::Tensor x, torch::Tensor buf);
+void bn_bwd_reduce_into(torch::Tensor x, torch::Tensor gy,
+                        torch::Tensor mean, torch::Tensor rstd,
+                        torch::Tensor buf);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> bn_finalize(
+    torch::Tensor buf, double count, double eps);
 std::tuple<torch::Tensor, torch::Tensor> bn_bwd_reduce(torch::Tensor x,
                                                        torch::Tensor gy,
                                                        torch::Tensor mean,
@@ -48,6 +54,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "per-layer sum of squares (LARS)");
   m.def("lars_update", &lars_update, "per-layer trust-ratio update (LARS)");
   m.def("bn_stats", &bn_stats, "per-channel (sum, sumsq, count) (K6)");
+  m.def("bn_stats_into", &bn_stats_into,
+        "per-channel (sum|sumsq) into a caller buffer (K6)");
+  m.def("bn_bwd_reduce_into", &bn_bwd_reduce_into,
+        "per-channel (sum gy|sum gy*xhat) into a caller buffer (K6)");
+  m.def("bn_finalize", &bn_finalize,
+        "mean/rstd/var from the reduced stats buffer (one launch)");
   m.def("bn_fwd", &bn_fwd, "BN normalize with global stats (K7)");
   m.def("bn_bwd_reduce", &bn_bwd_reduce,
         "per-channel (sum gy, sum gy*xhat) (K6)");

@@ -99,7 +99,9 @@ __device__ __forceinline__ void load4(const T16* p, float (&r)[4]) {
 // CT threads per row (each owning 4 consecutive channels), RPB = 256/CT
 // rows in flight per block; per-thread fp32 accumulators; LDS tree over
 // the RPB row-threads; one atomicAdd per (block, channel).
-template <typename T, bool BWD>
+// CPT = channels per thread (4 or 8); 16-bit dtypes use uint2/uint4
+// (8/16 B) per row visit, fp32 uses float4/2xfloat4.
+template <typename T, bool BWD, int CPT>
 __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
                                     const T* __restrict__ gy,
                                     const float* __restrict__ mean,
@@ -107,19 +109,22 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
                                     float* __restrict__ out0,
                                     float* __restrict__ out1,
                                     int C, long M) {
-  __shared__ float lds[CMN_BLOCK * 8];  // 8 KiB: [tid][8]
-  const int ct = min(C / 4, (int)blockDim.x);
+  __shared__ float lds[CMN_BLOCK * 2 * CPT > 16384 ? 16384
+                                                   : CMN_BLOCK * 2 * CPT];
+  const int ct = min(C / CPT, (int)blockDim.x);
   const int rpb = blockDim.x / ct;
-  const int tile_ch = ct * 4;
+  const int tile_ch = ct * CPT;
   const int tc = threadIdx.x % ct;
   const int tr = threadIdx.x / ct;
   for (int c_base = 0; c_base < C; c_base += tile_ch) {
-    const int c = c_base + tc * 4;
-    float a[8] = {0, 0, 0, 0, 0, 0, 0, 0};  // 4×(sum0, sum1)
-    float m4[4] = {0, 0, 0, 0}, r4[4] = {0, 0, 0, 0};
+    const int c = c_base + tc * CPT;
+    float a[2 * CPT];
+#pragma unroll
+    for (int k = 0; k < 2 * CPT; ++k) a[k] = 0.f;
+    float m4[CPT], r4[CPT];
     if (BWD) {
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
+      for (int k = 0; k < CPT; ++k) {
         m4[k] = mean[c + k];
         r4[k] = rstd[c + k];
       }
@@ -128,11 +133,14 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
     if (tr < rpb) {
       for (long m = blockIdx.x * (long)rpb + tr; m < M; m += row_stride) {
         const long idx = m * C + c;
-        float xv[4], gv[4];
-        load4(x + idx, xv);
-        if (BWD) load4(gy + idx, gv);
+        float xv[CPT], gv[CPT];
 #pragma unroll
-        for (int k = 0; k < 4; ++k) {
+        for (int q = 0; q < CPT / 4; ++q) {
+          load4(x + idx + 4 * q, *(float(*)[4])(xv + 4 * q));
+          if (BWD) load4(gy + idx + 4 * q, *(float(*)[4])(gv + 4 * q));
+        }
+#pragma unroll
+        for (int k = 0; k < CPT; ++k) {
           if (BWD) {
             a[2 * k] += gv[k];
             a[2 * k + 1] += gv[k] * ((xv[k] - m4[k]) * r4[k]);
@@ -144,22 +152,26 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
       }
     }
 #pragma unroll
-    for (int k = 0; k < 8; ++k) lds[threadIdx.x * 8 + k] = a[k];
+    for (int k = 0; k < 2 * CPT; ++k)
+      lds[threadIdx.x * 2 * CPT + k] = a[k];
     __syncthreads();
-    // cross-row tree: threads with tr < off accumulate row tr+off
-    for (int off = rpb >> 1; off > 0; off >>= 1) {
-      if (tr < off) {
+    // cross-row tree (non-power-of-two-safe ceil halving)
+    for (int rem = rpb; rem > 1;) {
+      const int off = (rem + 1) >> 1;
+      if (tr < rem - off) {
 #pragma unroll
-        for (int k = 0; k < 8; ++k)
-          lds[threadIdx.x * 8 + k] += lds[(threadIdx.x + off * ct) * 8 + k];
+        for (int k = 0; k < 2 * CPT; ++k)
+          lds[threadIdx.x * 2 * CPT + k] +=
+              lds[(threadIdx.x + off * ct) * 2 * CPT + k];
       }
       __syncthreads();
+      rem = off;
     }
     if (tr == 0) {
 #pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        atomicAdd(&out0[c + k], lds[threadIdx.x * 8 + 2 * k]);
-        atomicAdd(&out1[c + k], lds[threadIdx.x * 8 + 2 * k + 1]);
+      for (int k = 0; k < CPT; ++k) {
+        atomicAdd(&out0[c + k], lds[threadIdx.x * 2 * CPT + 2 * k]);
+        atomicAdd(&out1[c + k], lds[threadIdx.x * 2 * CPT + 2 * k + 1]);
       }
     }
     __syncthreads();
@@ -171,6 +183,14 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
 // MODE 1: gx = γ·rstd·(gy − (x̂·sgx + sg)/M)
 // Vectorized: 8 elems/lane.  CHANNELS_LAST → the 8 elems are 8 consecutive
 // channels (requires C % 8 == 0); NCHW → same channel (requires S % 8 == 0).
+__device__ __forceinline__ void loadp8(const float* p, int c0,
+                                       float (&r)[8]) {
+  const float4 a = *reinterpret_cast<const float4*>(p + c0);
+  const float4 b = *reinterpret_cast<const float4*>(p + c0 + 4);
+  r[0] = a.x; r[1] = a.y; r[2] = a.z; r[3] = a.w;
+  r[4] = b.x; r[5] = b.y; r[6] = b.z; r[7] = b.w;
+}
+
 template <typename T, int MODE, bool CHANNELS_LAST>
 __global__ void bn_pointwise_vec_kernel(const T* __restrict__ x,
                                         const T* __restrict__ gy,
@@ -190,16 +210,36 @@ __global__ void bn_pointwise_vec_kernel(const T* __restrict__ x,
     float xv[8], gv[8], o[8];
     V8<T>::load(x + base, xv);
     if (MODE == 1) V8<T>::load(gy + base, gv);
+    // ONE channel computation per 8-vector: channels-last vectors cover 8
+    // consecutive channels (C % 8 == 0), NCHW vectors share one channel.
+    float mn[8], rs[8], gm[8], bt[8], s0[8], s1[8];
+    if (CHANNELS_LAST) {
+      const int c0 = (int)(base % (long)C);
+      loadp8(mean, c0, mn);
+      loadp8(rstd, c0, rs);
+      loadp8(gamma, c0, gm);
+      if (MODE == 0) {
+        loadp8(beta, c0, bt);
+      } else {
+        loadp8(sg, c0, s0);
+        loadp8(sgx, c0, s1);
+      }
+    } else {
+      const int c = (int)((base / S) % C);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        mn[k] = mean[c]; rs[k] = rstd[c]; gm[k] = gamma[c];
+        if (MODE == 0) bt[k] = beta[c];
+        else { s0[k] = sg[c]; s1[k] = sgx[c]; }
+      }
+    }
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      const int c = CHANNELS_LAST ? (int)((base + k) % C)
-                                  : (int)((base / S) % C);
-      const float xh = (xv[k] - mean[c]) * rstd[c];
+      const float xh = (xv[k] - mn[k]) * rs[k];
       if (MODE == 0) {
-        o[k] = xh * gamma[c] + beta[c];
+        o[k] = xh * gm[k] + bt[k];
       } else {
-        o[k] = gamma[c] * rstd[c]
-               * (gv[k] - (xh * sgx[c] + sg[c]) * inv_m);
+        o[k] = gm[k] * rs[k] * (gv[k] - (xh * s1[k] + s0[k]) * inv_m);
       }
     }
     V8<T>::store(out + base, o);
@@ -229,6 +269,24 @@ __global__ void bn_pointwise_kernel(const T* __restrict__ x,
       const float g = float(gy[i]);
       out[i] = T(gamma[c] * rstd[c] * (g - (xh * sgx[c] + sg[c]) * inv_m));
     }
+  }
+}
+
+// mean/var/rstd from the (all-reduced) [sum | sqsum | count?] buffer in
+// ONE tiny launch (replaces ~6 C-sized torch ops per BN call).
+__global__ void bn_finalize_kernel(const float* __restrict__ buf,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ rstd,
+                                   float* __restrict__ var,
+                                   float count, float eps, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const float inv = 1.f / (count > 0.f ? count : buf[2 * C]);
+  if (c < C) {
+    const float mu = buf[c] * inv;
+    const float v = fmaxf(buf[C + c] * inv - mu * mu, 0.f);
+    mean[c] = mu;
+    var[c] = v;
+    rstd[c] = rsqrtf(v + eps);
   }
 }
 
@@ -264,13 +322,23 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
     auto run = [&](auto bwd_tag) {
       constexpr bool IS_BWD = decltype(bwd_tag)::value;
       if (cl && C % 4 == 0) {
-        const int ct = std::min<long>(C / 4, CMN_BLOCK);
+        const int cpt = (C % 8 == 0) ? 8 : 4;
+        const int ct = std::min<long>(C / cpt, CMN_BLOCK);
         const int rpb = CMN_BLOCK / ct;
-        const int grid = (int)std::min<long>((M + rpb - 1) / rpb, 312);
-        hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD>),
-                           dim3(grid), dim3(CMN_BLOCK), 0, stream.stream(),
-                           xp, gp, mp, rp, out0.data_ptr<float>(),
-                           out1.data_ptr<float>(), C, M);
+        const int grid = (int)std::min<long>((M + rpb - 1) / rpb, 1024);
+        if (cpt == 8) {
+          hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD, 8>),
+                             dim3(grid), dim3(CMN_BLOCK), 0,
+                             stream.stream(), xp, gp, mp, rp,
+                             out0.data_ptr<float>(),
+                             out1.data_ptr<float>(), C, M);
+        } else {
+          hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD, 4>),
+                             dim3(grid), dim3(CMN_BLOCK), 0,
+                             stream.stream(), xp, gp, mp, rp,
+                             out0.data_ptr<float>(),
+                             out1.data_ptr<float>(), C, M);
+        }
       } else if (cl) {
         // C not divisible by 4: per-channel scalar blocks (cold path)
         hipLaunchKernelGGL((bn_reduce_nchw_kernel<scalar_t, IS_BWD>),
@@ -293,6 +361,40 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
     if (bwd) run(std::true_type{});
     else run(std::false_type{});
   });
+}
+
+void bn_stats_into(torch::Tensor x, torch::Tensor buf) {
+  TORCH_CHECK(x.is_cuda() && buf.scalar_type() == at::kFloat);
+  const int C = (int)x.size(1);
+  TORCH_CHECK(buf.numel() >= 2 * C);
+  auto out0 = buf.narrow(0, 0, C);
+  auto out1 = buf.narrow(0, C, C);
+  bn_reduce(x, nullptr, nullptr, nullptr, out0, out1);
+}
+
+void bn_bwd_reduce_into(torch::Tensor x, torch::Tensor gy,
+                        torch::Tensor mean, torch::Tensor rstd,
+                        torch::Tensor buf) {
+  const int C = (int)x.size(1);
+  TORCH_CHECK(buf.numel() >= 2 * C);
+  auto out0 = buf.narrow(0, 0, C);
+  auto out1 = buf.narrow(0, C, C);
+  bn_reduce(x, &gy, &mean, &rstd, out0, out1);
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> bn_finalize(
+    torch::Tensor buf, double count, double eps) {
+  const int C = (int)(count > 0 ? buf.numel() / 2 : (buf.numel() - 1) / 2);
+  auto opts = torch::TensorOptions().dtype(at::kFloat).device(buf.device());
+  auto mean = torch::empty({C}, opts);
+  auto rstd = torch::empty({C}, opts);
+  auto var = torch::empty({C}, opts);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                     0, stream.stream(), buf.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     var.data_ptr<float>(), (float)count, (float)eps, C);
+  return {mean, rstd, var};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, long> bn_stats(torch::Tensor x) {

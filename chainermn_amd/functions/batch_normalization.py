@@ -38,19 +38,24 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, comm, eps):
         C = x.shape[1]
+        buf = torch.empty(2 * C + 1, dtype=torch.float32, device=x.device)
         if x.is_cuda:
-            s, sq, count = ops.require().bn_stats(x)
+            ext = ops.require()
+            ext.bn_stats_into(x, buf)
+            buf[2 * C] = float(x.numel() // C)
+            buf = comm.allreduce(buf)
+            mean, rstd, var = ext.bn_finalize(buf, -1.0, eps)
+            total = buf[2 * C]
         else:
             s, sq, count = _local_sums(x)
-        buf = torch.empty(2 * C + 1, dtype=torch.float32, device=x.device)
-        buf[:C] = s
-        buf[C:2 * C] = sq
-        buf[2 * C] = float(count)
-        buf = comm.allreduce(buf)
-        total = buf[2 * C]
-        mean = buf[:C] / total
-        var = buf[C:2 * C] / total - mean * mean
-        rstd = torch.rsqrt(var.clamp_min(0) + eps)
+            buf[:C] = s
+            buf[C:2 * C] = sq
+            buf[2 * C] = float(count)
+            buf = comm.allreduce(buf)
+            total = buf[2 * C]
+            mean = buf[:C] / total
+            var = buf[C:2 * C] / total - mean * mean
+            rstd = torch.rsqrt(var.clamp_min(0) + eps)
         if x.is_cuda:
             y = ops.require().bn_fwd(x, mean, rstd, gamma.float(),
                                      beta.float())
@@ -68,7 +73,10 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
         comm = ctx.comm
         C = x.shape[1]
         if x.is_cuda:
-            gbeta, ggamma = ops.require().bn_bwd_reduce(x, gy, mean, rstd)
+            buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+            ops.require().bn_bwd_reduce_into(x, gy, mean, rstd, buf)
+            gbeta = buf[:C]
+            ggamma = buf[C:]
         else:
             dims = [d for d in range(x.dim()) if d != 1]
             gyf = gy.float()

@@ -28,10 +28,11 @@ class _FusedBatchNorm(torch.autograd.Function):
     def forward(ctx, x, gamma, beta, eps):
         if x.is_cuda:
             ext = ops.require()
-            s, sq, count = ext.bn_stats(x)
-            mean = s / count
-            var = (sq / count - mean * mean).clamp_min_(0)
-            rstd = torch.rsqrt(var + eps)
+            C = x.shape[1]
+            count = x.numel() // C
+            buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+            ext.bn_stats_into(x, buf)
+            mean, rstd, var = ext.bn_finalize(buf, float(count), eps)
             y = ext.bn_fwd(x, mean, rstd, gamma.float(), beta.float())
         else:
             dims = [d for d in range(x.dim()) if d != 1]
@@ -53,7 +54,11 @@ class _FusedBatchNorm(torch.autograd.Function):
         count = ctx.count
         if x.is_cuda:
             ext = ops.require()
-            gbeta, ggamma = ext.bn_bwd_reduce(x, gy, mean, rstd)
+            C = x.shape[1]
+            buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+            ext.bn_bwd_reduce_into(x, gy, mean, rstd, buf)
+            gbeta = buf[:C]
+            ggamma = buf[C:]
             gx = ext.bn_bwd_input(x, gy, mean, rstd, gamma.float(), gbeta,
                                   ggamma, float(count))
         else:
