@@ -106,8 +106,7 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
                                     const T* __restrict__ gy,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ rstd,
-                                    float* __restrict__ out0,
-                                    float* __restrict__ out1,
+                                    float* __restrict__ partials,
                                     int C, long M) {
   __shared__ float lds[CMN_BLOCK * 2 * CPT > 16384 ? 16384
                                                    : CMN_BLOCK * 2 * CPT];
@@ -168,14 +167,31 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
       rem = off;
     }
     if (tr == 0) {
+      // per-block partials: no same-address atomic chains (the serialized
+      // atomicAdd per channel cost ~2x kernel time at grid 1024)
 #pragma unroll
       for (int k = 0; k < CPT; ++k) {
-        atomicAdd(&out0[c + k], lds[threadIdx.x * 2 * CPT + 2 * k]);
-        atomicAdd(&out1[c + k], lds[threadIdx.x * 2 * CPT + 2 * k + 1]);
+        partials[blockIdx.x * 2 * C + c + k] =
+            lds[threadIdx.x * 2 * CPT + 2 * k];
+        partials[blockIdx.x * 2 * C + C + c + k] =
+            lds[threadIdx.x * 2 * CPT + 2 * k + 1];
       }
     }
     __syncthreads();
   }
+}
+
+// stage 2: out[j] = sum over blocks of partials[g][j], j in [0, 2C)
+__global__ void bn_reduce_stage2_kernel(const float* __restrict__ partials,
+                                        float* __restrict__ out0,
+                                        float* __restrict__ out1,
+                                        int C, int n_blocks) {
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= 2 * C) return;
+  float acc = 0.f;
+  for (int g = 0; g < n_blocks; ++g) acc += partials[(long)g * 2 * C + j];
+  if (j < C) out0[j] = acc;
+  else out1[j - C] = acc;
 }
 
 // ---------- elementwise normalize / input-grad ---------------------------
@@ -326,19 +342,23 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
         const int ct = std::min<long>(C / cpt, CMN_BLOCK);
         const int rpb = CMN_BLOCK / ct;
         const int grid = (int)std::min<long>((M + rpb - 1) / rpb, 1024);
+        auto partials = torch::empty({grid, 2L * C},
+            torch::TensorOptions().dtype(at::kFloat).device(x.device()));
+        float* pp = partials.data_ptr<float>();
         if (cpt == 8) {
           hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD, 8>),
                              dim3(grid), dim3(CMN_BLOCK), 0,
-                             stream.stream(), xp, gp, mp, rp,
-                             out0.data_ptr<float>(),
-                             out1.data_ptr<float>(), C, M);
+                             stream.stream(), xp, gp, mp, rp, pp, C, M);
         } else {
           hipLaunchKernelGGL((bn_reduce_cl_kernel<scalar_t, IS_BWD, 4>),
                              dim3(grid), dim3(CMN_BLOCK), 0,
-                             stream.stream(), xp, gp, mp, rp,
-                             out0.data_ptr<float>(),
-                             out1.data_ptr<float>(), C, M);
+                             stream.stream(), xp, gp, mp, rp, pp, C, M);
         }
+        hipLaunchKernelGGL(bn_reduce_stage2_kernel,
+                           dim3((2 * C + CMN_BLOCK - 1) / CMN_BLOCK),
+                           dim3(CMN_BLOCK), 0, stream.stream(), pp,
+                           out0.data_ptr<float>(), out1.data_ptr<float>(),
+                           C, grid);
       } else if (cl) {
         // C not divisible by 4: per-channel scalar blocks (cold path)
         hipLaunchKernelGGL((bn_reduce_nchw_kernel<scalar_t, IS_BWD>),
