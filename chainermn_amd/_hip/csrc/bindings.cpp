@@ -29,7 +29,8 @@ void bn_bwd_reduce_into(torch::Tensor x, torch::Tensor gy,
                         torch::Tensor mean, torch::Tensor rstd,
                         torch::Tensor buf);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> bn_finalize(
-    torch::Tensor buf, double count, double eps);
+    torch::Tensor buf, double count, double eps, double momentum,
+    double unbias, torch::Tensor r_mean, torch::Tensor r_var);
 std::tuple<torch::Tensor, torch::Tensor> bn_bwd_reduce(torch::Tensor x,
                                                        torch::Tensor gy,
                                                        torch::Tensor mean,
@@ -59,7 +60,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce_into", &bn_bwd_reduce_into,
         "per-channel (sum gy|sum gy*xhat) into a caller buffer (K6)");
   m.def("bn_finalize", &bn_finalize,
-        "mean/rstd/var from the reduced stats buffer (one launch)");
+        "mean/rstd/var from the reduced stats buffer, optional fused "
+        "running-stat update (one launch)",
+        py::arg("buf"), py::arg("count"), py::arg("eps"),
+        py::arg("momentum") = 0.0, py::arg("unbias") = 1.0,
+        py::arg("r_mean") = torch::Tensor(),
+        py::arg("r_var") = torch::Tensor());
   m.def("bn_fwd", &bn_fwd, "BN normalize with global stats (K7)");
   m.def("bn_bwd_reduce", &bn_bwd_reduce,
         "per-channel (sum gy, sum gy*xhat) (K6)");

@@ -181,7 +181,10 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
   }
 }
 
-// stage 2: out[j] = sum over blocks of partials[g][j], j in [0, 2C)
+// stage 2: out[j] = sum over blocks of partials[g][j], j in [0, 2C).
+// 2-D grid: x covers channels, y splits the partial blocks (so the sum
+// parallelizes; a single-block serial loop here cost 235 us) -> at most
+// gridDim.y atomics per output word (out zeroed by caller).
 __global__ void bn_reduce_stage2_kernel(const float* __restrict__ partials,
                                         float* __restrict__ out0,
                                         float* __restrict__ out1,
@@ -189,9 +192,11 @@ __global__ void bn_reduce_stage2_kernel(const float* __restrict__ partials,
   const int j = blockIdx.x * blockDim.x + threadIdx.x;
   if (j >= 2 * C) return;
   float acc = 0.f;
-  for (int g = 0; g < n_blocks; ++g) acc += partials[(long)g * 2 * C + j];
-  if (j < C) out0[j] = acc;
-  else out1[j - C] = acc;
+  for (int g = blockIdx.y; g < n_blocks; g += gridDim.y)
+    acc += partials[(long)g * 2 * C + j];
+  float* dst = (j < C) ? &out0[j] : &out1[j - C];
+  if (gridDim.y == 1) *dst = acc;
+  else atomicAdd(dst, acc);
 }
 
 // ---------- elementwise normalize / input-grad ---------------------------
@@ -294,6 +299,9 @@ __global__ void bn_finalize_kernel(const float* __restrict__ buf,
                                    float* __restrict__ mean,
                                    float* __restrict__ rstd,
                                    float* __restrict__ var,
+                                   float* __restrict__ r_mean,
+                                   float* __restrict__ r_var,
+                                   float momentum, float unbias,
                                    float count, float eps, int C) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   const float inv = 1.f / (count > 0.f ? count : buf[2 * C]);
@@ -303,6 +311,12 @@ __global__ void bn_finalize_kernel(const float* __restrict__ buf,
     mean[c] = mu;
     var[c] = v;
     rstd[c] = rsqrtf(v + eps);
+    if (r_mean != nullptr) {
+      // fused running-stat update (the separate torch ops were 2 mixed-
+      // dtype kernel launches per BN layer)
+      r_mean[c] = (1.f - momentum) * r_mean[c] + momentum * mu;
+      r_var[c] = (1.f - momentum) * r_var[c] + momentum * v * unbias;
+    }
   }
 }
 
@@ -354,8 +368,9 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
                              dim3(grid), dim3(CMN_BLOCK), 0,
                              stream.stream(), xp, gp, mp, rp, pp, C, M);
         }
+        const int gsplit = std::max(1, std::min(grid / 8, 32));
         hipLaunchKernelGGL(bn_reduce_stage2_kernel,
-                           dim3((2 * C + CMN_BLOCK - 1) / CMN_BLOCK),
+                           dim3((2 * C + CMN_BLOCK - 1) / CMN_BLOCK, gsplit),
                            dim3(CMN_BLOCK), 0, stream.stream(), pp,
                            out0.data_ptr<float>(), out1.data_ptr<float>(),
                            C, grid);
@@ -403,17 +418,28 @@ void bn_bwd_reduce_into(torch::Tensor x, torch::Tensor gy,
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> bn_finalize(
-    torch::Tensor buf, double count, double eps) {
+    torch::Tensor buf, double count, double eps, double momentum,
+    double unbias, torch::Tensor r_mean, torch::Tensor r_var) {
   const int C = (int)(count > 0 ? buf.numel() / 2 : (buf.numel() - 1) / 2);
   auto opts = torch::TensorOptions().dtype(at::kFloat).device(buf.device());
   auto mean = torch::empty({C}, opts);
   auto rstd = torch::empty({C}, opts);
   auto var = torch::empty({C}, opts);
+  float* rm = nullptr;
+  float* rv = nullptr;
+  if (r_mean.defined() && r_mean.numel()) {
+    TORCH_CHECK(r_mean.scalar_type() == at::kFloat
+                && r_var.scalar_type() == at::kFloat,
+                "running stats must be fp32");
+    rm = r_mean.data_ptr<float>();
+    rv = r_var.data_ptr<float>();
+  }
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
                      0, stream.stream(), buf.data_ptr<float>(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     var.data_ptr<float>(), (float)count, (float)eps, C);
+                     var.data_ptr<float>(), rm, rv, (float)momentum,
+                     (float)unbias, (float)count, (float)eps, C);
   return {mean, rstd, var};
 }
 

@@ -25,14 +25,20 @@ def _shape(t, x):
 class _FusedBatchNorm(torch.autograd.Function):
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps):
+    def forward(ctx, x, gamma, beta, eps, momentum=0.0, r_mean=None,
+                r_var=None):
         if x.is_cuda:
             ext = ops.require()
             C = x.shape[1]
             count = x.numel() // C
             buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
             ext.bn_stats_into(x, buf)
-            mean, rstd, var = ext.bn_finalize(buf, float(count), eps)
+            if r_mean is not None:
+                unbias = count / max(count - 1, 1)
+                mean, rstd, var = ext.bn_finalize(
+                    buf, float(count), eps, momentum, unbias, r_mean, r_var)
+            else:
+                mean, rstd, var = ext.bn_finalize(buf, float(count), eps)
             y = ext.bn_fwd(x, mean, rstd, gamma.float(), beta.float())
         else:
             dims = [d for d in range(x.dim()) if d != 1]
@@ -70,7 +76,8 @@ class _FusedBatchNorm(torch.autograd.Function):
             gx = (_shape(gamma.float() * rstd, x)
                   * (gyf - (xhat * _shape(ggamma, x)
                             + _shape(gbeta, x)) / count)).to(x.dtype)
-        return gx, ggamma.to(gamma.dtype), gbeta.to(gamma.dtype), None
+        return (gx, ggamma.to(gamma.dtype), gbeta.to(gamma.dtype),
+                None, None, None, None)
 
 
 class FusedBatchNorm2d(nn.Module):
@@ -106,11 +113,29 @@ class FusedBatchNorm2d(nn.Module):
         beta = self.bias if self.bias is not None else torch.zeros(
             self.num_features, device=x.device, dtype=x.dtype)
         if self.training or not self.track_running_stats:
+            if self.track_running_stats and x.is_cuda:
+                # running stats stay fp32 even when the model is cast to
+                # bf16 (a bf16 buffer would force mixed-dtype update
+                # kernels AND destroy the stats' precision)
+                if self.running_mean.dtype != torch.float32:
+                    self.running_mean.data = self.running_mean.data.float()
+                    self.running_var.data = self.running_var.data.float()
+                m = self.momentum if self.momentum is not None else 0.1
+                y, mean, var = _FusedBatchNorm.apply(
+                    x, gamma, beta, self.eps, m, self.running_mean,
+                    self.running_var)
+                with torch.no_grad():
+                    self.num_batches_tracked += 1
+                return y
             y, mean, var = _FusedBatchNorm.apply(x, gamma, beta, self.eps)
             if self.track_running_stats:
                 with torch.no_grad():
                     n = x.numel() // x.shape[1]
                     m = self.momentum if self.momentum is not None else 0.1
+                    rm = self.running_mean
+                    if rm.dtype != torch.float32:
+                        self.running_mean.data = rm.data.float()
+                        self.running_var.data = self.running_var.data.float()
                     self.running_mean.mul_(1 - m).add_(mean.detach(),
                                                        alpha=m)
                     unbiased = var.detach() * (n / max(n - 1, 1))
@@ -118,7 +143,7 @@ class FusedBatchNorm2d(nn.Module):
                     self.num_batches_tracked += 1
             return y
         # eval mode: normalize with running stats
-        rstd = torch.rsqrt(self.running_var + self.eps)
+        rstd = torch.rsqrt(self.running_var.float() + self.eps)
         if x.is_cuda:
             return ops.require().bn_fwd(x, self.running_mean.float(), rstd,
                                         gamma.float(), beta.float())
