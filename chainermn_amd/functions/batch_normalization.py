@@ -44,7 +44,9 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
             ext.bn_stats_into(x, buf)
             buf[2 * C] = float(x.numel() // C)
             buf = comm.allreduce(buf)
-            mean, rstd, var = ext.bn_finalize(buf, -1.0, eps)
+            mean, rstd, var = ext.bn_finalize(buf, -1.0, eps, 0.0, 1.0,
+                                              buf.new_empty(0),
+                                              buf.new_empty(0))
             total = buf[2 * C]
         else:
             s, sq, count = _local_sums(x)

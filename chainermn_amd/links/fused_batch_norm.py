@@ -38,7 +38,9 @@ class _FusedBatchNorm(torch.autograd.Function):
                 mean, rstd, var = ext.bn_finalize(
                     buf, float(count), eps, momentum, unbias, r_mean, r_var)
             else:
-                mean, rstd, var = ext.bn_finalize(buf, float(count), eps)
+                mean, rstd, var = ext.bn_finalize(buf, float(count), eps, 0.0,
+                                                  1.0, buf.new_empty(0),
+                                                  buf.new_empty(0))
             y = ext.bn_fwd(x, mean, rstd, gamma.float(), beta.float())
         else:
             dims = [d for d in range(x.dim()) if d != 1]
