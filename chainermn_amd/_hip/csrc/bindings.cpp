@@ -28,6 +28,20 @@ void bn_stats_into(torch::Tensor x, torch::Tensor buf);
 void bn_bwd_reduce_into(torch::Tensor x, torch::Tensor gy,
                         torch::Tensor mean, torch::Tensor rstd,
                         torch::Tensor buf);
+void bn_bwd_reduce_act_into(torch::Tensor x, torch::Tensor gy,
+                            torch::Tensor mean, torch::Tensor rstd,
+                            torch::Tensor gamma, torch::Tensor beta,
+                            torch::Tensor yout, long relu_mode,
+                            torch::Tensor buf);
+torch::Tensor bn_fwd_act(torch::Tensor x, torch::Tensor mean,
+                         torch::Tensor rstd, torch::Tensor gamma,
+                         torch::Tensor beta, torch::Tensor res,
+                         long relu_mode);
+std::tuple<torch::Tensor, torch::Tensor> bn_bwd_input_act(
+    torch::Tensor x, torch::Tensor gy, torch::Tensor mean,
+    torch::Tensor rstd, torch::Tensor gamma, torch::Tensor beta,
+    torch::Tensor gbeta, torch::Tensor ggamma, torch::Tensor yout,
+    long relu_mode, bool want_gres, double total_count);
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> bn_finalize(
     torch::Tensor buf, double count, double eps, double momentum,
     double unbias, torch::Tensor r_mean, torch::Tensor r_var);
@@ -70,4 +84,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce,
         "per-channel (sum gy, sum gy*xhat) (K6)");
   m.def("bn_bwd_input", &bn_bwd_input, "BN input gradient (K7)");
+  m.def("bn_bwd_reduce_act_into", &bn_bwd_reduce_act_into,
+        "BN backward reduce with fused relu mask");
+  m.def("bn_fwd_act", &bn_fwd_act,
+        "y = act(bn(x) [+ res]) (fused relu / residual add)");
+  m.def("bn_bwd_input_act", &bn_bwd_input_act,
+        "BN input grad with fused relu mask [+ gres]");
 }

@@ -4,3 +4,4 @@ from chainermn_amd.links.multi_node_chain_list import MultiNodeChainList  # noqa
 from chainermn_amd.links.n_step_rnn import create_multi_node_n_step_rnn  # noqa: F401
 from chainermn_amd.links.n_step_rnn import MultiNodeNStepRNN  # noqa: F401
 from chainermn_amd.links.fused_batch_norm import FusedBatchNorm2d  # noqa: F401
+from chainermn_amd.links.fused_batch_norm import FusedBNReLU2d  # noqa: F401

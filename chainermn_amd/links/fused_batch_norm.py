@@ -1,13 +1,23 @@
-"""FusedBatchNorm2d — local (per-rank) BatchNorm on the CDNA4 BN kernels.
+"""Fused BatchNorm (+ReLU, +Add+ReLU) on the CDNA4 BN kernels.
 
-Drop-in for ``torch.nn.BatchNorm2d``.  Exists because torch's native
+Drop-ins for ``torch.nn.BatchNorm2d`` and for the BN→ReLU /
+BN→add-residual→ReLU motifs of ResNet.  Exists because torch's native
 channels_last BatchNorm kernels cost 66% of a ResNet-50 training step on
-MI355X (see profiles/): its collect_statistics/backward_reduce kernels run
-~20x above the bandwidth bound.  This module runs the same four fused
-passes as MultiNodeBatchNormalization (bn_stats / bn_fwd / bn_bwd_reduce /
-bn_bwd_input in _hip/csrc/mnbn.hip) without the cross-rank all-reduce —
-the data-parallel configs of the reference used per-rank local BN
-(global-batch BN is the separate MNBN link).
+MI355X (see profiles/), and the separate ReLU / residual-add elementwise
+kernels cost several more ms of pure HBM re-reads.  The fusions keep the
+activation in the producing kernel:
+
+* ``FusedBatchNorm2d``      : y = bn(x)
+* ``FusedBNReLU2d``         : y = relu(bn(x)) — backward recomputes the
+  relu mask from gamma*xhat+beta (zero extra tensor traffic)
+* ``FusedBatchNorm2d.forward_add_relu(x, res)`` :
+  y = relu(bn(x) + res) — backward masks with the saved output y and
+  emits the residual gradient from the same kernel pass
+
+Kernels: _hip/csrc/mnbn.hip (shared with MultiNodeBatchNormalization).
+Running statistics are updated inside the bn_finalize kernel and always
+kept fp32 (a bf16-cast model would otherwise hold bf16 stat buffers and
+pay a mixed-dtype update kernel per layer per step).
 """
 
 import torch
@@ -22,15 +32,22 @@ def _shape(t, x):
     return t.view(s)
 
 
-class _FusedBatchNorm(torch.autograd.Function):
+def _empty(x):
+    return x.new_empty(0)
+
+
+class _FusedBNAct(torch.autograd.Function):
+    """relu_mode: 0 = bn, 1 = bn+relu (affine mask), 2 = bn+add+relu."""
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps, momentum=0.0, r_mean=None,
-                r_var=None):
+    def forward(ctx, x, res, gamma, beta, eps, relu_mode, momentum,
+                r_mean, r_var):
+        C = x.shape[1]
+        count = x.numel() // C
+        gamma_f = gamma.float()
+        beta_f = beta.float()
         if x.is_cuda:
             ext = ops.require()
-            C = x.shape[1]
-            count = x.numel() // C
             buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
             ext.bn_stats_into(x, buf)
             if r_mean is not None:
@@ -38,53 +55,89 @@ class _FusedBatchNorm(torch.autograd.Function):
                 mean, rstd, var = ext.bn_finalize(
                     buf, float(count), eps, momentum, unbias, r_mean, r_var)
             else:
-                mean, rstd, var = ext.bn_finalize(buf, float(count), eps, 0.0,
-                                                  1.0, buf.new_empty(0),
-                                                  buf.new_empty(0))
-            y = ext.bn_fwd(x, mean, rstd, gamma.float(), beta.float())
+                mean, rstd, var = ext.bn_finalize(
+                    buf, float(count), eps, 0.0, 1.0, _empty(buf),
+                    _empty(buf))
+            y = ext.bn_fwd_act(x, mean, rstd, gamma_f, beta_f,
+                               res if res is not None else _empty(x),
+                               1 if relu_mode else 0)
         else:
             dims = [d for d in range(x.dim()) if d != 1]
             xf = x.float()
-            count = x.numel() // x.shape[1]
             mean = xf.mean(dim=dims)
             var = xf.var(dim=dims, unbiased=False)
             rstd = torch.rsqrt(var + eps)
-            y = ((xf - _shape(mean, x)) * _shape(rstd, x)
-                 * _shape(gamma.float(), x)
-                 + _shape(beta.float(), x)).to(x.dtype)
-        ctx.save_for_backward(x, gamma, mean, rstd)
+            y = (xf - _shape(mean, x)) * _shape(rstd, x) \
+                * _shape(gamma_f, x) + _shape(beta_f, x)
+            if res is not None:
+                y = y + res.float()
+            if relu_mode:
+                y = torch.relu(y)
+            y = y.to(x.dtype)
+            if r_mean is not None:
+                with torch.no_grad():
+                    unbias = count / max(count - 1, 1)
+                    r_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                    r_var.mul_(1 - momentum).add_(var * unbias,
+                                                  alpha=momentum)
+        if relu_mode == 2:
+            ctx.save_for_backward(x, gamma, beta, mean, rstd, y)
+        else:
+            ctx.save_for_backward(x, gamma, beta, mean, rstd)
+        ctx.relu_mode = relu_mode
         ctx.count = count
-        return y, mean, var
+        ctx.has_res = res is not None
+        return y
 
     @staticmethod
-    def backward(ctx, gy, g_mean, g_var):
-        x, gamma, mean, rstd = ctx.saved_tensors
+    def backward(ctx, gy):
+        relu_mode = ctx.relu_mode
         count = ctx.count
+        if relu_mode == 2:
+            x, gamma, beta, mean, rstd, yout = ctx.saved_tensors
+        else:
+            x, gamma, beta, mean, rstd = ctx.saved_tensors
+            yout = None
+        gamma_f = gamma.float()
+        beta_f = beta.float()
+        C = x.shape[1]
         if x.is_cuda:
             ext = ops.require()
-            C = x.shape[1]
             buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
-            ext.bn_bwd_reduce_into(x, gy, mean, rstd, buf)
+            ext.bn_bwd_reduce_act_into(
+                x, gy, mean, rstd, gamma_f, beta_f,
+                yout if yout is not None else _empty(x), relu_mode, buf)
             gbeta = buf[:C]
             ggamma = buf[C:]
-            gx = ext.bn_bwd_input(x, gy, mean, rstd, gamma.float(), gbeta,
-                                  ggamma, float(count))
+            gx, gres = ext.bn_bwd_input_act(
+                x, gy, mean, rstd, gamma_f, beta_f, gbeta, ggamma,
+                yout if yout is not None else _empty(x), relu_mode,
+                ctx.has_res, float(count))
         else:
             dims = [d for d in range(x.dim()) if d != 1]
-            gyf = gy.float()
             xhat = (x.float() - _shape(mean, x)) * _shape(rstd, x)
+            gyf = gy.float()
+            if relu_mode == 1:
+                mask = (xhat * _shape(gamma_f, x) + _shape(beta_f, x)) > 0
+                gyf = gyf * mask
+            elif relu_mode == 2:
+                gyf = gyf * (yout.float() > 0)
             gbeta = gyf.sum(dim=dims)
             ggamma = (gyf * xhat).sum(dim=dims)
-            gx = (_shape(gamma.float() * rstd, x)
+            gx = (_shape(gamma_f * rstd, x)
                   * (gyf - (xhat * _shape(ggamma, x)
                             + _shape(gbeta, x)) / count)).to(x.dtype)
-        return (gx, ggamma.to(gamma.dtype), gbeta.to(gamma.dtype),
-                None, None, None, None)
+            gres = gyf.to(x.dtype) if ctx.has_res else None
+        g_res = gres if ctx.has_res else None
+        return (gx, g_res, ggamma.to(gamma.dtype), gbeta.to(beta.dtype),
+                None, None, None, None, None)
 
 
 class FusedBatchNorm2d(nn.Module):
     """torch.nn.BatchNorm2d-compatible (momentum/eps/affine semantics,
     unbiased running_var) on the fused gfx950 kernels."""
+
+    _relu_mode = 0
 
     def __init__(self, num_features, eps=1e-5, momentum=0.1, affine=True,
                  track_running_stats=True):
@@ -109,50 +162,66 @@ class FusedBatchNorm2d(nn.Module):
             self.register_buffer('running_mean', None)
             self.register_buffer('running_var', None)
 
-    def forward(self, x):
+    def _gamma_beta(self, x):
         gamma = self.weight if self.weight is not None else torch.ones(
             self.num_features, device=x.device, dtype=x.dtype)
         beta = self.bias if self.bias is not None else torch.zeros(
             self.num_features, device=x.device, dtype=x.dtype)
-        if self.training or not self.track_running_stats:
-            if self.track_running_stats and x.is_cuda:
-                # running stats stay fp32 even when the model is cast to
-                # bf16 (a bf16 buffer would force mixed-dtype update
-                # kernels AND destroy the stats' precision)
-                if self.running_mean.dtype != torch.float32:
-                    self.running_mean.data = self.running_mean.data.float()
-                    self.running_var.data = self.running_var.data.float()
-                m = self.momentum if self.momentum is not None else 0.1
-                y, mean, var = _FusedBatchNorm.apply(
-                    x, gamma, beta, self.eps, m, self.running_mean,
-                    self.running_var)
-                with torch.no_grad():
-                    self.num_batches_tracked += 1
-                return y
-            y, mean, var = _FusedBatchNorm.apply(x, gamma, beta, self.eps)
-            if self.track_running_stats:
-                with torch.no_grad():
-                    n = x.numel() // x.shape[1]
-                    m = self.momentum if self.momentum is not None else 0.1
-                    rm = self.running_mean
-                    if rm.dtype != torch.float32:
-                        self.running_mean.data = rm.data.float()
-                        self.running_var.data = self.running_var.data.float()
-                    self.running_mean.mul_(1 - m).add_(mean.detach(),
-                                                       alpha=m)
-                    unbiased = var.detach() * (n / max(n - 1, 1))
-                    self.running_var.mul_(1 - m).add_(unbiased, alpha=m)
-                    self.num_batches_tracked += 1
-            return y
-        # eval mode: normalize with running stats
+        return gamma, beta
+
+    def _running(self):
+        if not self.track_running_stats:
+            return None, None, 0.0
+        if self.running_mean.dtype != torch.float32:
+            # keep stats fp32 even when the model is cast to bf16
+            self.running_mean.data = self.running_mean.data.float()
+            self.running_var.data = self.running_var.data.float()
+        m = self.momentum if self.momentum is not None else 0.1
+        return self.running_mean, self.running_var, m
+
+    def _train_forward(self, x, res, relu_mode):
+        gamma, beta = self._gamma_beta(x)
+        r_mean, r_var, m = self._running()
+        y = _FusedBNAct.apply(x, res, gamma, beta, self.eps, relu_mode, m,
+                              r_mean, r_var)
+        if self.track_running_stats:
+            self.num_batches_tracked += 1
+        return y
+
+    def _eval_forward(self, x, res, relu_mode):
+        gamma, beta = self._gamma_beta(x)
         rstd = torch.rsqrt(self.running_var.float() + self.eps)
         if x.is_cuda:
-            return ops.require().bn_fwd(x, self.running_mean.float(), rstd,
-                                        gamma.float(), beta.float())
-        y = ((x.float() - _shape(self.running_mean, x)) * _shape(rstd, x)
-             * _shape(gamma.float(), x) + _shape(beta.float(), x))
+            return ops.require().bn_fwd_act(
+                x, self.running_mean.float(), rstd, gamma.float(),
+                beta.float(), res if res is not None else x.new_empty(0),
+                1 if relu_mode else 0)
+        y = (x.float() - _shape(self.running_mean.float(), x)) \
+            * _shape(rstd, x) * _shape(gamma.float(), x) \
+            + _shape(beta.float(), x)
+        if res is not None:
+            y = y + res.float()
+        if relu_mode:
+            y = torch.relu(y)
         return y.to(x.dtype)
+
+    def forward(self, x):
+        if self.training or not self.track_running_stats:
+            return self._train_forward(x, None, self._relu_mode)
+        return self._eval_forward(x, None, self._relu_mode)
+
+    def forward_add_relu(self, x, res):
+        """y = relu(bn(x) + res) in one kernel pass (ResNet residual)."""
+        if self.training or not self.track_running_stats:
+            return self._train_forward(x, res, 2)
+        return self._eval_forward(x, res, 2)
 
     def extra_repr(self):
         return '{num_features}, eps={eps}, momentum={momentum}, ' \
                'affine={affine}'.format(**self.__dict__)
+
+
+class FusedBNReLU2d(FusedBatchNorm2d):
+    """y = relu(bn(x)) — the relu never touches HBM separately."""
+
+    _relu_mode = 1

@@ -17,22 +17,36 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, in_ch, width, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d):
+                 norm_layer=nn.BatchNorm2d, fused_ops=False):
         super().__init__()
         out_ch = width * self.expansion
+        self.fused_ops = fused_ops
+        if fused_ops:
+            from chainermn_amd.links.fused_batch_norm import (
+                FusedBatchNorm2d, FusedBNReLU2d)
+            bn_relu, bn_plain = FusedBNReLU2d, FusedBatchNorm2d
+        else:
+            bn_relu = bn_plain = norm_layer
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = norm_layer(width)
+        self.bn1 = bn_relu(width)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = norm_layer(width)
+        self.bn2 = bn_relu(width)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = norm_layer(out_ch)
+        self.bn3 = bn_plain(out_ch)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x):
         identity = x
+        if self.fused_ops:
+            # BN+ReLU and BN+Add+ReLU are single CDNA4 kernels
+            out = self.bn1(self.conv1(x))
+            out = self.bn2(self.conv2(out))
+            if self.downsample is not None:
+                identity = self.downsample(x)
+            return self.bn3.forward_add_relu(self.conv3(out), identity)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
@@ -45,13 +59,22 @@ class Bottleneck(nn.Module):
 class ResNet50(nn.Module):
 
     def __init__(self, num_classes=1000, norm_layer=nn.BatchNorm2d,
-                 zero_init_residual=True):
+                 zero_init_residual=True, fused_ops=False):
         super().__init__()
+        if fused_ops:
+            from chainermn_amd.links.fused_batch_norm import (
+                FusedBatchNorm2d, FusedBNReLU2d)
+            self._stem_bn = FusedBNReLU2d
+            norm_layer = FusedBatchNorm2d
+        else:
+            self._stem_bn = norm_layer
         self._norm = norm_layer
+        self._fused_ops = fused_ops
         self.in_ch = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = norm_layer(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = self._stem_bn(64)
+        self.relu = (nn.Identity() if fused_ops
+                     else nn.ReLU(inplace=True))
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, 3, stride=1)
         self.layer2 = self._make_layer(128, 4, stride=2)
@@ -80,10 +103,11 @@ class ResNet50(nn.Module):
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
                 norm_layer(out_ch))
         layers = [Bottleneck(self.in_ch, width, stride, downsample,
-                             norm_layer)]
+                             norm_layer, fused_ops=self._fused_ops)]
         self.in_ch = out_ch
         for _ in range(1, blocks):
-            layers.append(Bottleneck(out_ch, width, norm_layer=norm_layer))
+            layers.append(Bottleneck(out_ch, width, norm_layer=norm_layer,
+                                     fused_ops=self._fused_ops))
         return nn.Sequential(*layers)
 
     def forward(self, x):

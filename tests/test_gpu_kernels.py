@@ -373,3 +373,66 @@ def test_fused_batch_norm_vs_torch(dtype, cl):
     ye = bn(x32.to(dtype))
     yre = ref(x32)
     assert torch.allclose(ye.float(), yre, atol=tol)
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_bn_relu_add_gpu(dtype):
+    from chainermn_amd.links import FusedBatchNorm2d, FusedBNReLU2d
+    torch.manual_seed(11)
+    N, C, H, W = 8, 32, 14, 14
+    x32 = torch.randn(N, C, H, W, device=DEV)
+    res32 = torch.randn(N, C, H, W, device=DEV)
+    gy32 = torch.randn(N, C, H, W, device=DEV)
+    to = lambda t: t.to(dtype).to(memory_format=torch.channels_last)
+    tol = 5e-2 if dtype == torch.bfloat16 else 1e-4
+    # bn+relu
+    x = to(x32).requires_grad_(True)
+    xr = x32.clone().requires_grad_(True)
+    bn = FusedBNReLU2d(C).to(DEV)
+    ref = torch.nn.BatchNorm2d(C).to(DEV)
+    y = bn(x)
+    yr = torch.relu(ref(xr))
+    y.backward(to(gy32))
+    yr.backward(gy32)
+    assert torch.allclose(y.float(), yr.detach(), atol=tol)
+    assert torch.allclose(x.grad.float(), xr.grad, atol=tol)
+    assert torch.allclose(bn.weight.grad.float(), ref.weight.grad,
+                          atol=tol * 20, rtol=1e-2)
+    # bn+add+relu
+    x2 = to(x32).requires_grad_(True)
+    r2 = to(res32).requires_grad_(True)
+    xr2 = x32.clone().requires_grad_(True)
+    rr2 = res32.clone().requires_grad_(True)
+    bn2 = FusedBatchNorm2d(C).to(DEV)
+    ref2 = torch.nn.BatchNorm2d(C).to(DEV)
+    y2 = bn2.forward_add_relu(x2, r2)
+    yr2 = torch.relu(ref2(xr2) + rr2)
+    y2.backward(to(gy32))
+    yr2.backward(gy32)
+    assert torch.allclose(y2.float(), yr2.detach(), atol=tol)
+    assert torch.allclose(x2.grad.float(), xr2.grad, atol=tol)
+    assert torch.allclose(r2.grad.float(), rr2.grad, atol=tol)
+
+
+def test_resnet_fused_ops_gpu_loss_decreases():
+    import chainermn_amd
+    from chainermn_amd.models import ResNet50
+    from chainermn_amd.parallel import FusedDataParallelSGD
+    comm = chainermn_amd.create_communicator('dummy')
+    torch.manual_seed(3)
+    model = ResNet50(num_classes=100, fused_ops=True).to(
+        device=DEV, dtype=torch.bfloat16,
+        memory_format=torch.channels_last)
+    opt = FusedDataParallelSGD(model, comm, lr=0.05, momentum=0.9,
+                               overlap=False)
+    x = torch.randn(16, 3, 64, 64, device=DEV, dtype=torch.bfloat16
+                    ).to(memory_format=torch.channels_last)
+    y = torch.randint(0, 100, (16,), device=DEV)
+    losses = []
+    for _ in range(12):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x).float(), y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, losses
