@@ -201,20 +201,30 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
   }
 }
 
-// stage 2: out[j] = Σ_g partials[g][j].  2-D grid: x covers channels, y
-// splits the partial blocks (a single serial block cost 235 us).
+// stage 2: out[j] = Σ_g partials[g][j].  256 threads = 4 g-lanes × 64
+// j-lanes (coalesced 256 B rows), LDS tree over the g-lanes — no atomics,
+// no pre-zeroed output (a serial single block cost 235 us; same-address
+// atomics with a split grid need zeroing and serialize).
 __global__ void bn_reduce_stage2_kernel(const float* __restrict__ partials,
                                         float* __restrict__ out0,
                                         float* __restrict__ out1,
                                         int C, int n_blocks) {
-  const int j = blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= 2 * C) return;
+  __shared__ float lds[CMN_BLOCK];
+  const int tj = threadIdx.x % 64;
+  const int tg = threadIdx.x / 64;
+  const int j = blockIdx.x * 64 + tj;
   float acc = 0.f;
-  for (int g = blockIdx.y; g < n_blocks; g += gridDim.y)
-    acc += partials[(long)g * 2 * C + j];
-  float* dst = (j < C) ? &out0[j] : &out1[j - C];
-  if (gridDim.y == 1) *dst = acc;
-  else atomicAdd(dst, acc);
+  if (j < 2 * C) {
+    for (int g = tg; g < n_blocks; g += 4)
+      acc += partials[(long)g * 2 * C + j];
+  }
+  lds[threadIdx.x] = acc;
+  __syncthreads();
+  if (tg == 0 && j < 2 * C) {
+    acc = lds[tj] + lds[tj + 64] + lds[tj + 128] + lds[tj + 192];
+    if (j < C) out0[j] = acc;
+    else out1[j - C] = acc;
+  }
 }
 
 // ---- pointwise -----------------------------------------------------------
@@ -406,12 +416,10 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
                              stream.stream(), xp, gp, yp, mp, rp, gmp, btp,
                              pp, C, M);
         }
-        const int gsplit = std::max(1, std::min(grid / 8, 32));
         hipLaunchKernelGGL(bn_reduce_stage2_kernel,
-                           dim3((2 * C + CMN_BLOCK - 1) / CMN_BLOCK, gsplit),
-                           dim3(CMN_BLOCK), 0, stream.stream(), pp,
-                           out0.data_ptr<float>(), out1.data_ptr<float>(),
-                           C, grid);
+                           dim3((2 * C + 63) / 64), dim3(CMN_BLOCK), 0,
+                           stream.stream(), pp, out0.data_ptr<float>(),
+                           out1.data_ptr<float>(), C, grid);
       } else {
         out0.zero_();
         out1.zero_();
