@@ -52,6 +52,12 @@ __device__ __forceinline__ void loadp8(const float* p, int c0,
   r[0] = a.x; r[1] = a.y; r[2] = a.z; r[3] = a.w;
   r[4] = b.x; r[5] = b.y; r[6] = b.z; r[7] = b.w;
 }
+template <typename T16>
+__device__ __forceinline__ void loadp8(const T16* p, int c0,
+                                       float (&r)[8]) {
+  load4(p + c0, *(float(*)[4])(r));
+  load4(p + c0 + 4, *(float(*)[4])(r + 4));
+}
 
 // ---- NCHW reduce (scalar fallback; bench path is channels-last) ----------
 template <typename T, bool BWD, int RM>
@@ -60,8 +66,8 @@ __global__ void bn_reduce_nchw_kernel(const T* __restrict__ x,
                                       const T* __restrict__ yout,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ rstd,
-                                      const float* __restrict__ gamma,
-                                      const float* __restrict__ beta,
+                                      const T* __restrict__ gamma,
+                                      const T* __restrict__ beta,
                                       float* __restrict__ out0,
                                       float* __restrict__ out1,
                                       int C, long N, long S) {
@@ -72,8 +78,8 @@ __global__ void bn_reduce_nchw_kernel(const T* __restrict__ x,
   float a0 = 0.f, a1 = 0.f;
   const float mu = BWD ? mean[c] : 0.f;
   const float rs = BWD ? rstd[c] : 0.f;
-  const float gm = (BWD && RM == 1) ? gamma[c] : 0.f;
-  const float bt = (BWD && RM == 1) ? beta[c] : 0.f;
+  const float gm = (BWD && RM == 1) ? float(gamma[c]) : 0.f;
+  const float bt = (BWD && RM == 1) ? float(beta[c]) : 0.f;
   for (long i = blockIdx.y * (long)blockDim.x + threadIdx.x; i < per;
        i += stride) {
     const long n = i / S;
@@ -118,8 +124,8 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
                                     const T* __restrict__ yout,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ rstd,
-                                    const float* __restrict__ gamma,
-                                    const float* __restrict__ beta,
+                                    const T* __restrict__ gamma,
+                                    const T* __restrict__ beta,
                                     float* __restrict__ partials,
                                     int C, long M) {
   __shared__ float lds[CMN_BLOCK * 2 * CPT > 16384 ? 16384
@@ -140,7 +146,10 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
       for (int k = 0; k < CPT; ++k) {
         m4[k] = mean[c + k];
         r4[k] = rstd[c + k];
-        if (RM == 1) { g4[k] = gamma[c + k]; b4[k] = beta[c + k]; }
+        if (RM == 1) {
+          g4[k] = float(gamma[c + k]);
+          b4[k] = float(beta[c + k]);
+        }
       }
     }
     const long row_stride = (long)gridDim.x * rpb;
@@ -201,22 +210,43 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
   }
 }
 
-// stage 2: out[j] = Σ_g partials[g][j].  256 threads = 4 g-lanes × 64
-// j-lanes (coalesced 256 B rows), LDS tree over the g-lanes — no atomics,
-// no pre-zeroed output (a serial single block cost 235 us; same-address
-// atomics with a split grid need zeroing and serialize).
-__global__ void bn_reduce_stage2_kernel(const float* __restrict__ partials,
-                                        float* __restrict__ out0,
-                                        float* __restrict__ out1,
-                                        int C, int n_blocks) {
+// stage 2 (hierarchical, no atomics, no pre-zeroed output — a serial
+// single block cost 235 us; same-address atomics serialize).
+// 2a: grid (jblocks, GS) — block (jb, gi) sums its g-slice -> mid[gi][2C]
+// 2b: grid (jblocks) — 4 g-lanes × 64 j-lanes + LDS tree over GS=16 mids
+#define BN_GS 16
+__global__ void bn_reduce_stage2a_kernel(const float* __restrict__ partials,
+                                         float* __restrict__ mid,
+                                         int C, int n_blocks) {
+  __shared__ float lds[CMN_BLOCK];
+  const int tj = threadIdx.x % 64;
+  const int tg = threadIdx.x / 64;
+  const int j = blockIdx.x * 64 + tj;
+  const int gi = blockIdx.y;
+  float acc = 0.f;
+  if (j < 2 * C) {
+    for (int g = gi * 4 + tg; g < n_blocks; g += BN_GS * 4)
+      acc += partials[(long)g * 2 * C + j];
+  }
+  lds[threadIdx.x] = acc;
+  __syncthreads();
+  if (tg == 0 && j < 2 * C)
+    mid[(long)gi * 2 * C + j] =
+        lds[tj] + lds[tj + 64] + lds[tj + 128] + lds[tj + 192];
+}
+
+__global__ void bn_reduce_stage2b_kernel(const float* __restrict__ mid,
+                                         float* __restrict__ out0,
+                                         float* __restrict__ out1, int C) {
   __shared__ float lds[CMN_BLOCK];
   const int tj = threadIdx.x % 64;
   const int tg = threadIdx.x / 64;
   const int j = blockIdx.x * 64 + tj;
   float acc = 0.f;
   if (j < 2 * C) {
-    for (int g = tg; g < n_blocks; g += 4)
-      acc += partials[(long)g * 2 * C + j];
+#pragma unroll
+    for (int g = tg; g < BN_GS; g += 4)
+      acc += mid[(long)g * 2 * C + j];
   }
   lds[threadIdx.x] = acc;
   __syncthreads();
@@ -239,8 +269,8 @@ __global__ void bn_pointwise_vec_kernel(const T* __restrict__ x,
                                         T* __restrict__ gres,
                                         const float* __restrict__ mean,
                                         const float* __restrict__ rstd,
-                                        const float* __restrict__ gamma,
-                                        const float* __restrict__ beta,
+                                        const T* __restrict__ gamma,
+                                        const T* __restrict__ beta,
                                         const float* __restrict__ sg,
                                         const float* __restrict__ sgx,
                                         float inv_m, int C, long S,
@@ -303,8 +333,8 @@ __global__ void bn_pointwise_kernel(const T* __restrict__ x,
                                     T* __restrict__ gres,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ rstd,
-                                    const float* __restrict__ gamma,
-                                    const float* __restrict__ beta,
+                                    const T* __restrict__ gamma,
+                                    const T* __restrict__ beta,
                                     const float* __restrict__ sg,
                                     const float* __restrict__ sgx,
                                     float inv_m, int C, long S, long total) {
@@ -314,16 +344,18 @@ __global__ void bn_pointwise_kernel(const T* __restrict__ x,
     const int c = CHANNELS_LAST ? (int)(i % C) : (int)((i / S) % C);
     const float xh = (float(x[i]) - mean[c]) * rstd[c];
     if (MODE == 0) {
-      float y = xh * gamma[c] + beta[c];
+      float y = xh * float(gamma[c]) + float(beta[c]);
       if (RES) y += float(aux[i]);
       if (RM == 1) y = fmaxf(y, 0.f);
       out[i] = T(y);
     } else {
       float g = float(gy[i]);
-      if (RM == 1) g = (xh * gamma[c] + beta[c]) > 0.f ? g : 0.f;
+      if (RM == 1) g = (xh * float(gamma[c]) + float(beta[c])) > 0.f
+                        ? g : 0.f;
       if (RM == 2) g = float(aux[i]) > 0.f ? g : 0.f;
       if (RES) gres[i] = T(g);
-      out[i] = T(gamma[c] * rstd[c] * (g - (xh * sgx[c] + sg[c]) * inv_m));
+      out[i] = T(float(gamma[c]) * rstd[c]
+                 * (g - (xh * sgx[c] + sg[c]) * inv_m));
     }
   }
 }
@@ -392,8 +424,10 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
     const scalar_t* yp = yout ? (const scalar_t*)yout->data_ptr() : nullptr;
     const float* mp = bwd ? mean->data_ptr<float>() : nullptr;
     const float* rp = bwd ? rstd->data_ptr<float>() : nullptr;
-    const float* gmp = gamma ? gamma->data_ptr<float>() : nullptr;
-    const float* btp = beta ? beta->data_ptr<float>() : nullptr;
+    const scalar_t* gmp = gamma
+        ? (const scalar_t*)gamma->data_ptr() : nullptr;
+    const scalar_t* btp = beta
+        ? (const scalar_t*)beta->data_ptr() : nullptr;
     auto run = [&](auto bwd_tag, auto rm_tag) {
       constexpr bool IS_BWD = decltype(bwd_tag)::value;
       constexpr int RM = decltype(rm_tag)::value;
@@ -416,10 +450,17 @@ static void bn_reduce(const torch::Tensor& x, const torch::Tensor* gy,
                              stream.stream(), xp, gp, yp, mp, rp, gmp, btp,
                              pp, C, M);
         }
-        hipLaunchKernelGGL(bn_reduce_stage2_kernel,
-                           dim3((2 * C + 63) / 64), dim3(CMN_BLOCK), 0,
-                           stream.stream(), pp, out0.data_ptr<float>(),
-                           out1.data_ptr<float>(), C, grid);
+        auto mid = torch::empty({BN_GS, 2L * C},
+            torch::TensorOptions().dtype(at::kFloat).device(x.device()));
+        const int jblocks = (2 * C + 63) / 64;
+        hipLaunchKernelGGL(bn_reduce_stage2a_kernel,
+                           dim3(jblocks, BN_GS), dim3(CMN_BLOCK), 0,
+                           stream.stream(), pp, mid.data_ptr<float>(), C,
+                           grid);
+        hipLaunchKernelGGL(bn_reduce_stage2b_kernel, dim3(jblocks),
+                           dim3(CMN_BLOCK), 0, stream.stream(),
+                           mid.data_ptr<float>(), out0.data_ptr<float>(),
+                           out1.data_ptr<float>(), C);
       } else {
         out0.zero_();
         out1.zero_();
@@ -470,7 +511,8 @@ static void bn_pointwise(int mode, const torch::Tensor& x,
     const scalar_t* ap = aux ? (const scalar_t*)aux->data_ptr() : nullptr;
     scalar_t* op = (scalar_t*)out.data_ptr();
     scalar_t* grp = gres ? (scalar_t*)gres->data_ptr() : nullptr;
-    const float* btp = beta ? beta->data_ptr<float>() : nullptr;
+    const scalar_t* btp = beta
+        ? (const scalar_t*)beta->data_ptr() : nullptr;
     const float* sgp = sg ? sg->data_ptr<float>() : nullptr;
     const float* sgxp = sgx ? sgx->data_ptr<float>() : nullptr;
     auto launch = [&](auto mode_tag, auto cl_tag, auto rm_tag,
@@ -484,15 +526,15 @@ static void bn_pointwise(int mode, const torch::Tensor& x,
             (bn_pointwise_vec_kernel<scalar_t, MODE, CL, RM, RES>),
             dim3(cmn_grid(total / 8, CMN_BLOCK)), dim3(CMN_BLOCK), 0,
             stream.stream(), xp, gp, ap, op, grp, mean.data_ptr<float>(),
-            rstd.data_ptr<float>(), gamma.data_ptr<float>(), btp, sgp,
-            sgxp, inv_m, C, S, total / 8);
+            rstd.data_ptr<float>(), (const scalar_t*)gamma.data_ptr(),
+            btp, sgp, sgxp, inv_m, C, S, total / 8);
       } else {
         hipLaunchKernelGGL(
             (bn_pointwise_kernel<scalar_t, MODE, CL, RM, RES>),
             dim3(cmn_grid(total, CMN_BLOCK)), dim3(CMN_BLOCK), 0,
             stream.stream(), xp, gp, ap, op, grp, mean.data_ptr<float>(),
-            rstd.data_ptr<float>(), gamma.data_ptr<float>(), btp, sgp,
-            sgxp, inv_m, C, S, total);
+            rstd.data_ptr<float>(), (const scalar_t*)gamma.data_ptr(),
+            btp, sgp, sgxp, inv_m, C, S, total);
       }
     };
     using i0 = std::integral_constant<int, 0>;
@@ -546,8 +588,7 @@ void bn_bwd_reduce_act_into(torch::Tensor x, torch::Tensor gy,
 void bn_bwd_reduce_into(torch::Tensor x, torch::Tensor gy,
                         torch::Tensor mean, torch::Tensor rstd,
                         torch::Tensor buf) {
-  bn_bwd_reduce_act_into(x, gy, mean, rstd, mean, mean, torch::Tensor(), 0,
-                         buf);
+  bn_bwd_reduce_act_into(x, gy, mean, rstd, x, x, torch::Tensor(), 0, buf);
 }
 
 std::tuple<torch::Tensor, torch::Tensor, long> bn_stats(torch::Tensor x) {

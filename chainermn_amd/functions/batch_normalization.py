@@ -59,8 +59,8 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
             var = buf[C:2 * C] / total - mean * mean
             rstd = torch.rsqrt(var.clamp_min(0) + eps)
         if x.is_cuda:
-            y = ops.require().bn_fwd(x, mean, rstd, gamma.float(),
-                                     beta.float())
+            y = ops.require().bn_fwd(x, mean, rstd, gamma.to(x.dtype),
+                                     beta.to(x.dtype))
         else:
             xhat = (x.float() - _bcast_shape(mean, x)) * _bcast_shape(rstd, x)
             y = (xhat * _bcast_shape(gamma.float(), x)
@@ -92,7 +92,8 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
         gbeta_g = buf[:C]
         ggamma_g = buf[C:]
         if x.is_cuda:
-            gx = ops.require().bn_bwd_input(x, gy, mean, rstd, gamma.float(),
+            gx = ops.require().bn_bwd_input(x, gy, mean, rstd,
+                                            gamma.to(x.dtype),
                                             gbeta_g, ggamma_g, float(total))
         else:
             gyf = gy.float()

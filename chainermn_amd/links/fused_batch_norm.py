@@ -44,8 +44,8 @@ class _FusedBNAct(torch.autograd.Function):
                 r_mean, r_var):
         C = x.shape[1]
         count = x.numel() // C
-        gamma_f = gamma.float()
-        beta_f = beta.float()
+        gamma_f = gamma.to(x.dtype)
+        beta_f = beta.to(x.dtype)
         if x.is_cuda:
             ext = ops.require()
             buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
@@ -68,7 +68,7 @@ class _FusedBNAct(torch.autograd.Function):
             var = xf.var(dim=dims, unbiased=False)
             rstd = torch.rsqrt(var + eps)
             y = (xf - _shape(mean, x)) * _shape(rstd, x) \
-                * _shape(gamma_f, x) + _shape(beta_f, x)
+                * _shape(gamma_f.float(), x) + _shape(beta_f.float(), x)
             if res is not None:
                 y = y + res.float()
             if relu_mode:
@@ -98,8 +98,8 @@ class _FusedBNAct(torch.autograd.Function):
         else:
             x, gamma, beta, mean, rstd = ctx.saved_tensors
             yout = None
-        gamma_f = gamma.float()
-        beta_f = beta.float()
+        gamma_f = gamma.to(x.dtype)
+        beta_f = beta.to(x.dtype)
         C = x.shape[1]
         if x.is_cuda:
             ext = ops.require()
@@ -118,13 +118,14 @@ class _FusedBNAct(torch.autograd.Function):
             xhat = (x.float() - _shape(mean, x)) * _shape(rstd, x)
             gyf = gy.float()
             if relu_mode == 1:
-                mask = (xhat * _shape(gamma_f, x) + _shape(beta_f, x)) > 0
+                mask = (xhat * _shape(gamma_f.float(), x)
+                        + _shape(beta_f.float(), x)) > 0
                 gyf = gyf * mask
             elif relu_mode == 2:
                 gyf = gyf * (yout.float() > 0)
             gbeta = gyf.sum(dim=dims)
             ggamma = (gyf * xhat).sum(dim=dims)
-            gx = (_shape(gamma_f * rstd, x)
+            gx = (_shape(gamma_f.float() * rstd, x)
                   * (gyf - (xhat * _shape(ggamma, x)
                             + _shape(gbeta, x)) / count)).to(x.dtype)
             gres = gyf.to(x.dtype) if ctx.has_res else None
@@ -193,8 +194,8 @@ class FusedBatchNorm2d(nn.Module):
         rstd = torch.rsqrt(self.running_var.float() + self.eps)
         if x.is_cuda:
             return ops.require().bn_fwd_act(
-                x, self.running_mean.float(), rstd, gamma.float(),
-                beta.float(), res if res is not None else x.new_empty(0),
+                x, self.running_mean.float(), rstd, gamma.to(x.dtype),
+                beta.to(x.dtype), res if res is not None else x.new_empty(0),
                 1 if relu_mode else 0)
         y = (x.float() - _shape(self.running_mean.float(), x)) \
             * _shape(rstd, x) * _shape(gamma.float(), x) \
