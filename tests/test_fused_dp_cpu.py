@@ -1,0 +1,67 @@
+"""FusedDataParallel logic on a 2-process gloo world (CPU, fp32 arenas,
+overlap=False): ranks stay identical and match the serial mean-grad
+update — insurance for the 8-GPU run the driver performs."""
+
+import torch
+import torch.nn as nn
+
+import chainermn_amd
+from chainermn_amd.parallel import FusedDataParallelSGD
+from chainermn_amd.testing import run_distributed
+
+
+def _model(seed):
+    torch.manual_seed(seed)
+    return nn.Sequential(nn.Linear(6, 16), nn.ReLU(), nn.Linear(16, 4))
+
+
+def _worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    model = _model(0)
+    opt = FusedDataParallelSGD(model, comm, lr=0.1, momentum=0.9,
+                               weight_decay=1e-4, overlap=False,
+                               grad_dtype=torch.float32)
+    for step in range(3):
+        opt.zero_grad()
+        torch.manual_seed(50 + step * size + rank)
+        x = torch.randn(5, 6)
+        model(x).pow(2).mean().backward()
+        opt.step()
+    return [p.detach().clone() for p in model.parameters()]
+
+
+def test_fused_dp_matches_serial():
+    results = run_distributed(_worker, 2)
+    results = [[torch.from_numpy(a) for a in r] for r in results]
+    for a, b in zip(results[0], results[1]):
+        assert torch.allclose(a, b, atol=1e-6)
+    # serial reference
+    ref = _model(0)
+    opt = torch.optim.SGD(ref.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+    for step in range(3):
+        opt.zero_grad()
+        losses = []
+        for rank in range(2):
+            torch.manual_seed(50 + step * 2 + rank)
+            x = torch.randn(5, 6)
+            losses.append(ref(x).pow(2).mean())
+        (sum(losses) / 2).backward()
+        opt.step()
+    for p, r in zip(results[0], ref.parameters()):
+        assert torch.allclose(p, r.detach(), atol=1e-5), \
+            (p - r.detach()).abs().max()
+
+
+def test_bench_cpu_world2():
+    """bench.py runs end-to-end on a 2-proc CPU world (mlp model)."""
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29617', 'bench.py', '--model', 'mlp',
+         '--steps', '3', '--warmup', '1', '--batch-size', '8'],
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert '"n_gpus": 2' in out.stdout, out.stdout
