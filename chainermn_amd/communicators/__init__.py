@@ -46,4 +46,16 @@ def create_communicator(communicator_name=None, **kwargs):
     if communicator_name not in table:
         raise ValueError('unknown communicator: {} (choose from {})'.format(
             communicator_name, sorted(table)))
-    return table[communicator_name](**kwargs)
+    cls = table[communicator_name]
+    # reference-factory behaviour: tuning kwargs that a communicator does
+    # not take (allreduce_grad_dtype / batched_copy on the non-pure_rccl
+    # classes) are accepted and ignored with a warning
+    import inspect
+    accepted = set(inspect.signature(cls.__init__).parameters)
+    dropped = [k for k in kwargs if k not in accepted]
+    if dropped:
+        import warnings
+        warnings.warn('{} ignores kwargs {}'.format(
+            communicator_name, dropped))
+        kwargs = {k: v for k, v in kwargs.items() if k in accepted}
+    return cls(**kwargs)

@@ -444,3 +444,42 @@ def test_resnet_fused_ops_gpu_loss_decreases():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0] * 0.7, losses
+
+
+def test_double_buffering_optimizer_gpu():
+    """Generic DoubleBufferingOptimizer (arena + side-stream cast path) on
+    one GPU: finite losses, params update after step 1, wait() drains."""
+    import chainermn_amd
+    import torch.nn as nn
+    comm = chainermn_amd.create_communicator('pure_rccl')
+    torch.manual_seed(5)
+    model = nn.Linear(16, 8).to(DEV)
+    opt = chainermn_amd.create_multi_node_optimizer(
+        torch.optim.SGD(model.parameters(), lr=0.1), comm,
+        double_buffering=True).setup(model)
+    w0 = model.weight.detach().clone()
+    for i in range(3):
+        opt.zero_grad()
+        loss = model(torch.randn(4, 16, device=DEV)).pow(2).mean()
+        loss.backward()
+        opt.step()
+    opt.wait()
+    assert torch.isfinite(model.weight).all()
+    assert not torch.allclose(model.weight, w0)
+
+
+def test_pure_rccl_mean_grad_single_rank():
+    """PureRcclCommunicator pack->allreduce->unpack on world_size=1."""
+    import chainermn_amd
+    import torch.nn as nn
+    comm = chainermn_amd.create_communicator(
+        'pure_rccl', allreduce_grad_dtype='bf16')
+    model = nn.Linear(33, 7).to(DEV)
+    for p in model.parameters():
+        p.grad = torch.randn_like(p)
+    before = [p.grad.clone() for p in model.parameters()]
+    comm.multi_node_mean_grad(model)
+    for p, b in zip(model.parameters(), before):
+        # world=1: value preserved through the bf16 wire round-trip
+        assert torch.allclose(p.grad, b.to(torch.bfloat16).float(),
+                              atol=1e-2)
