@@ -43,7 +43,8 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
             ext = ops.require()
             ext.bn_stats_into(x, buf)
             buf[2 * C] = float(x.numel() // C)
-            buf = comm.allreduce(buf)
+            if comm.size > 1:
+                buf = comm.allreduce(buf)
             mean, rstd, var = ext.bn_finalize(buf, -1.0, eps, 0.0, 1.0,
                                               buf.new_empty(0),
                                               buf.new_empty(0))
@@ -88,7 +89,8 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
         buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
         buf[:C] = gbeta
         buf[C:] = ggamma
-        buf = comm.allreduce(buf)
+        if comm.size > 1:
+            buf = comm.allreduce(buf)
         gbeta_g = buf[:C]
         ggamma_g = buf[C:]
         if x.is_cuda:

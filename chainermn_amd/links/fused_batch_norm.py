@@ -163,6 +163,16 @@ class FusedBatchNorm2d(nn.Module):
             self.register_buffer('running_mean', None)
             self.register_buffer('running_var', None)
 
+    def _apply(self, fn, recurse=True):
+        # num_batches_tracked stays a CPU scalar: incrementing a GPU long
+        # tensor costs one kernel launch per layer per step
+        ret = super()._apply(fn, recurse)
+        if self.num_batches_tracked is not None \
+                and self.num_batches_tracked.is_cuda:
+            self.num_batches_tracked.data = \
+                self.num_batches_tracked.data.cpu()
+        return ret
+
     def _gamma_beta(self, x):
         gamma = self.weight if self.weight is not None else torch.ones(
             self.num_features, device=x.device, dtype=x.dtype)

@@ -124,3 +124,47 @@ def test_seq2seq_model_parallel():
                           model.encoder.embed.weight.grad, atol=1e-5)
     assert torch.allclose(torch.from_numpy(results[1]['g']),
                           model.decoder.embed.weight.grad, atol=1e-5)
+
+
+def _nstep_rnn_worker(rank, size):
+    import torch.nn as nn
+    from chainermn_amd.links import create_multi_node_n_step_rnn
+    comm = chainermn_amd.create_communicator('naive')
+    torch.manual_seed(4)
+    rnn_a = nn.LSTM(6, 6, 1, batch_first=True)
+    rnn_b = nn.LSTM(6, 6, 1, batch_first=True)
+    torch.manual_seed(9)
+    x1 = torch.randn(2, 5, 6)
+    x2 = torch.randn(2, 4, 6)
+    if rank == 0:
+        wrapped = create_multi_node_n_step_rnn(rnn_a, comm, rank_in=None,
+                                               rank_out=1)
+        out, h_n, delegate = wrapped(x1)
+        delegate.backward()
+        return {'g': rnn_a.weight_ih_l0.grad}
+    else:
+        wrapped = create_multi_node_n_step_rnn(rnn_b, comm, rank_in=0,
+                                               rank_out=None)
+        out, h_n, _ = wrapped(x2)
+        out.sum().backward()
+        return {'out': out.detach(), 'g': rnn_b.weight_ih_l0.grad}
+
+
+def test_multi_node_n_step_rnn():
+    results = run_distributed(_nstep_rnn_worker, 2)
+    import torch.nn as nn
+    torch.manual_seed(4)
+    rnn_a = nn.LSTM(6, 6, 1, batch_first=True)
+    rnn_b = nn.LSTM(6, 6, 1, batch_first=True)
+    torch.manual_seed(9)
+    x1 = torch.randn(2, 5, 6)
+    x2 = torch.randn(2, 4, 6)
+    _, state = rnn_a(x1)
+    out, _ = rnn_b(x2, state)
+    out.sum().backward()
+    assert torch.allclose(torch.from_numpy(results[1]['out']), out.detach(),
+                          atol=1e-6)
+    assert torch.allclose(torch.from_numpy(results[0]['g']),
+                          rnn_a.weight_ih_l0.grad, atol=1e-5)
+    assert torch.allclose(torch.from_numpy(results[1]['g']),
+                          rnn_b.weight_ih_l0.grad, atol=1e-5)
