@@ -102,6 +102,8 @@ class Trainer(object):
         return self.epoch >= n
 
     def step(self, batch):
+        import time as _time
+        _t0 = _time.perf_counter()
         self.observation = {}
         model = self.model
         if self.loss_fn is not None:
@@ -118,6 +120,8 @@ class Trainer(object):
         self.observation.setdefault('main/loss', float(loss.detach()))
         self.observation['iteration'] = self.iteration
         self.observation['epoch'] = self.epoch
+        self.observation['step_time_ms'] = (
+            (_time.perf_counter() - _t0) * 1000)
         for e in self._extensions:
             if _match_trigger(self, e.trigger):
                 e.ext(self)
