@@ -193,6 +193,34 @@ class FusedDataParallelSGD(object):
             self.v.copy_(sd['v'])
 
 
+class FusedDataParallelAdam(FusedDataParallelSGD):
+    """Adam/AdamW over the flat arenas (one fused kernel per step)."""
+
+    def __init__(self, model, comm, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                 weight_decay=0.0, adamw=False, overlap=True,
+                 grad_dtype=torch.bfloat16):
+        super().__init__(model, comm, lr, momentum=0.0,
+                         weight_decay=weight_decay, overlap=overlap,
+                         grad_dtype=grad_dtype)
+        n = self.state.plan.total
+        dev = self.state.device
+        self.betas = betas
+        self.eps = eps
+        self.adamw = adamw
+        self.m = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.vv = torch.zeros(n, dtype=torch.float32, device=dev)
+        self._adam_step = 0
+
+    def _apply(self, g):
+        self._adam_step += 1
+        ops.fused_adam(self.state.master, g, self.m, self.vv, self.lr,
+                       self.betas[0], self.betas[1], self.eps,
+                       weight_decay=self.weight_decay,
+                       step=self._adam_step, adamw=self.adamw,
+                       grad_scale=1.0 / max(self.comm.size, 1),
+                       p_lowp=self.state.lowp)
+
+
 class FusedDataParallelLARS(FusedDataParallelSGD):
     """Layer-wise adaptive rate scaling (You et al. 2017) over the same
     arenas — the large-batch config (BASELINE.json config 5).  Per layer:
