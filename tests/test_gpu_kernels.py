@@ -483,3 +483,32 @@ def test_pure_rccl_mean_grad_single_rank():
         # world=1: value preserved through the bf16 wire round-trip
         assert torch.allclose(p.grad, b.to(torch.bfloat16).float(),
                               atol=1e-2)
+
+
+def test_fused_dp_adam_matches_eager():
+    import chainermn_amd
+    import torch.nn as nn
+    from chainermn_amd.parallel import FusedDataParallelAdam
+    comm = chainermn_amd.create_communicator('dummy')
+    torch.manual_seed(21)
+    model = nn.Sequential(nn.Linear(12, 24), nn.ReLU(),
+                          nn.Linear(24, 6)).to(DEV)
+    ref = nn.Sequential(nn.Linear(12, 24), nn.ReLU(),
+                        nn.Linear(24, 6)).to(DEV)
+    ref.load_state_dict(model.state_dict())
+    opt = FusedDataParallelAdam(model, comm, lr=1e-2, weight_decay=1e-2,
+                                adamw=True, overlap=False)
+    ropt = torch.optim.AdamW(ref.parameters(), lr=1e-2, weight_decay=1e-2)
+    for step in range(3):
+        torch.manual_seed(200 + step)
+        x = torch.randn(8, 12, device=DEV)
+        opt.zero_grad()
+        model(x).pow(2).mean().backward()
+        opt.step()
+        ropt.zero_grad()
+        ref(x).pow(2).mean().backward()
+        ropt.step()
+    torch.cuda.synchronize()
+    for p, r in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, r.data, atol=1e-5), \
+            (p.data - r.data).abs().max()
