@@ -92,18 +92,103 @@ class _FlatState(object):
                 self.master[off:off + sz].copy_(p.data.reshape(-1).float())
 
 
+class _BucketEngine(object):
+    """Zero-staleness comm/compute overlap: the flat grad arena is split
+    into contiguous buckets; per-param post-accumulate-grad hooks launch
+    each bucket's in-place all-reduce on the comm stream as soon as the
+    bucket's last gradient lands during backward (the modern alternative
+    to the reference's 1-step-stale double buffering).  Buckets are
+    launched in a FIXED order (a ready-prefix walk over reverse arena
+    order, which tracks backward's param order) so every rank issues RCCL
+    collectives identically — no cross-rank deadlock by construction.
+    """
+
+    def __init__(self, state, comm, bucket_mb=25):
+        self.state = state
+        self.comm = comm
+        arena = state.grad_arena()
+        esize = arena.element_size()
+        bucket_elems = max(int(bucket_mb * (1 << 20)) // esize, 1)
+        # buckets over slot ranges; params assigned by start offset
+        spans = list(zip(state.plan.offsets_cpu, state.plan.sizes_cpu))
+        buckets = []
+        cur_params, cur_start = [], None
+        for i, (off, sz) in enumerate(spans):
+            if cur_start is None:
+                cur_start = off
+            cur_params.append(i)
+            end = (off + sz + ops.ALIGN - 1) // ops.ALIGN * ops.ALIGN
+            if end - cur_start >= bucket_elems or i == len(spans) - 1:
+                buckets.append((cur_start, end, list(cur_params)))
+                cur_params, cur_start = [], None
+        # reverse arena order approximates backward completion order
+        self.buckets = list(reversed(buckets))
+        self.param_bucket = {}
+        for bi, (_, _, pidx) in enumerate(self.buckets):
+            for i in pidx:
+                self.param_bucket[id(state.params[i])] = bi
+        self.use_gpu = arena.is_cuda
+        if self.use_gpu:
+            self.stream = torch.cuda.Stream()
+            self.events = [torch.cuda.Event() for _ in self.buckets]
+        self._reset()
+        for p in state.params:
+            p.register_post_accumulate_grad_hook(self._hook)
+
+    def _reset(self):
+        self.pending = [len(b[2]) for b in self.buckets]
+        self.launched = 0  # buckets [0, launched) already issued
+
+    def _hook(self, p):
+        bi = self.param_bucket[id(p)]
+        self.pending[bi] -= 1
+        self._launch_ready_prefix()
+
+    def _launch_ready_prefix(self):
+        while self.launched < len(self.buckets) \
+                and self.pending[self.launched] == 0:
+            start, end, _ = self.buckets[self.launched]
+            shard = self.state.grad_arena()[start:end]
+            if self.comm.size > 1:
+                if self.use_gpu:
+                    main = torch.cuda.current_stream()
+                    self.stream.wait_stream(main)
+                    with torch.cuda.stream(self.stream):
+                        dist.all_reduce(shard, group=self.comm.group)
+                        self.events[self.launched].record(self.stream)
+                else:
+                    dist.all_reduce(shard, group=self.comm.group)
+            elif self.use_gpu:
+                self.events[self.launched].record(
+                    torch.cuda.current_stream())
+            self.launched += 1
+
+    def finish(self):
+        """Flush un-launched buckets, wait for all; call before update."""
+        for bi in range(self.launched, len(self.buckets)):
+            self.pending[bi] = 0
+        self._launch_ready_prefix()
+        if self.use_gpu:
+            main = torch.cuda.current_stream()
+            for ev in self.events:
+                main.wait_event(ev)
+        self._reset()
+
+
 class FusedDataParallelSGD(object):
     """SGD/MomentumSGD over flat arenas with overlapped bf16 all-reduce.
 
     ``overlap=True`` → double-buffered (1-step-stale) gradients: the
     all-reduce of step n runs on a side stream under the fwd/bwd of step
-    n+1; the first step applies no update.  ``overlap=False`` → reduce and
-    update synchronously (0 staleness).
+    n+1; the first step applies no update (the reference's flagship
+    double-buffering config).  ``overlap='bucketed'`` → zero-staleness:
+    per-bucket all-reduce overlapped WITHIN backward via grad hooks.
+    ``overlap=False`` → synchronous.
     """
 
     def __init__(self, model, comm, lr, momentum=0.9, weight_decay=0.0,
                  dampening=0.0, nesterov=False, overlap=True,
-                 grad_dtype=torch.bfloat16):
+                 grad_dtype=torch.bfloat16, bucket_mb=25):
         self.comm = comm
         self.lr = lr
         self.momentum = momentum
@@ -111,14 +196,19 @@ class FusedDataParallelSGD(object):
         self.dampening = dampening
         self.nesterov = nesterov
         self.overlap = overlap
+        double_buf = overlap is True
         self.state = _FlatState(model, comm,
                                 grad_dtype=grad_dtype,
-                                n_grad_buffers=2 if overlap else 1)
+                                n_grad_buffers=2 if double_buf else 1)
         n = self.state.plan.total
         self.v = torch.zeros(n, dtype=torch.float32,
                              device=self.state.device) \
             if momentum else None
-        if overlap:
+        self.buckets = None
+        if overlap == 'bucketed':
+            self.buckets = _BucketEngine(self.state, comm,
+                                         bucket_mb=bucket_mb)
+        elif double_buf:
             self.stream = torch.cuda.Stream()
             self.ev_ar = [torch.cuda.Event(), torch.cuda.Event()]
         self._step = 0
@@ -147,6 +237,13 @@ class FusedDataParallelSGD(object):
 
     def step(self):
         st = self.state
+        if self.overlap == 'bucketed':
+            self.buckets.finish()
+            g = st.grad_arena()
+            self._apply(g)
+            g.zero_()
+            self._step += 1
+            return
         if not self.overlap:
             g = st.grad_arena()
             if self.comm.size > 1:
@@ -171,7 +268,7 @@ class FusedDataParallelSGD(object):
         self._step += 1
 
     def wait(self):
-        if self.overlap and self._step > 0:
+        if self.overlap is True and self._step > 0:
             torch.cuda.current_stream().wait_event(
                 self.ev_ar[1 - self.state.cur])
 

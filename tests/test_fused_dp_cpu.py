@@ -65,3 +65,39 @@ def test_bench_cpu_world2():
         capture_output=True, text=True, timeout=240)
     assert out.returncode == 0, out.stderr[-2000:]
     assert '"n_gpus": 2' in out.stdout, out.stdout
+
+
+def _bucketed_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    model = _model(0)
+    opt = FusedDataParallelSGD(model, comm, lr=0.1, momentum=0.9,
+                               weight_decay=1e-4, overlap='bucketed',
+                               grad_dtype=torch.float32, bucket_mb=1)
+    for step in range(3):
+        opt.zero_grad()
+        torch.manual_seed(50 + step * size + rank)
+        x = torch.randn(5, 6)
+        model(x).pow(2).mean().backward()
+        opt.step()
+    return [p.detach().clone() for p in model.parameters()]
+
+
+def test_bucketed_overlap_matches_serial():
+    results = run_distributed(_bucketed_worker, 2)
+    results = [[torch.from_numpy(a) for a in r] for r in results]
+    for a, b in zip(results[0], results[1]):
+        assert torch.allclose(a, b, atol=1e-6)
+    ref = _model(0)
+    opt = torch.optim.SGD(ref.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+    for step in range(3):
+        opt.zero_grad()
+        losses = []
+        for rank in range(2):
+            torch.manual_seed(50 + step * 2 + rank)
+            x = torch.randn(5, 6)
+            losses.append(ref(x).pow(2).mean())
+        (sum(losses) / 2).backward()
+        opt.step()
+    for p, r in zip(results[0], ref.parameters()):
+        assert torch.allclose(p, r.detach(), atol=1e-5)

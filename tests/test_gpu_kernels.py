@@ -512,3 +512,31 @@ def test_fused_dp_adam_matches_eager():
     for p, r in zip(model.parameters(), ref.parameters()):
         assert torch.allclose(p.data, r.data, atol=1e-5), \
             (p.data - r.data).abs().max()
+
+
+def test_bucketed_overlap_gpu_single_rank():
+    import chainermn_amd
+    import torch.nn as nn
+    from chainermn_amd.parallel import FusedDataParallelSGD
+    comm = chainermn_amd.create_communicator('dummy')
+    torch.manual_seed(31)
+    model = nn.Sequential(nn.Linear(64, 128), nn.ReLU(),
+                          nn.Linear(128, 10)).to(DEV)
+    ref = nn.Sequential(nn.Linear(64, 128), nn.ReLU(),
+                        nn.Linear(128, 10)).to(DEV)
+    ref.load_state_dict(model.state_dict())
+    opt = FusedDataParallelSGD(model, comm, lr=0.1, momentum=0.9,
+                               overlap='bucketed', bucket_mb=1)
+    ropt = torch.optim.SGD(ref.parameters(), lr=0.1, momentum=0.9)
+    for step in range(3):
+        torch.manual_seed(300 + step)
+        x = torch.randn(8, 64, device=DEV)
+        opt.zero_grad()
+        model(x).pow(2).mean().backward()
+        opt.step()
+        ropt.zero_grad()
+        ref(x).pow(2).mean().backward()
+        ropt.step()
+    torch.cuda.synchronize()
+    for p, r in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, r.data, atol=1e-5)
