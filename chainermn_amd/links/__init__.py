@@ -5,3 +5,4 @@ from chainermn_amd.links.n_step_rnn import create_multi_node_n_step_rnn  # noqa:
 from chainermn_amd.links.n_step_rnn import MultiNodeNStepRNN  # noqa: F401
 from chainermn_amd.links.fused_batch_norm import FusedBatchNorm2d  # noqa: F401
 from chainermn_amd.links.fused_batch_norm import FusedBNReLU2d  # noqa: F401
+from chainermn_amd.links.batch_normalization import MultiNodeBNReLU2d  # noqa: F401

@@ -4,27 +4,35 @@ Reference: chainermn/links/batch_normalization.py (~L1-180): drop-in for
 the framework's BatchNormalization that computes batch statistics over the
 *global* (cross-rank) batch and maintains running statistics from them.
 Essential for large-batch training (BASELINE.json config 5: batch 8192 +
-LARS).  Here it is a drop-in for ``torch.nn.BatchNorm1d/2d/3d``.
+LARS).
+
+Implementation: the fused CDNA4 BN engine (fused_batch_norm._FusedBNAct)
+with a communicator attached — the (Σx, Σx², Σgy', Σgy'x̂) buffers go over
+RCCL between the local reduction and the finalize/input kernels.  Running
+statistics use the reference's convention (decay, biased batch variance).
+``MultiNodeBNReLU2d`` / ``forward_add_relu`` give the multi-node BN the
+same ReLU / residual fusions as the local FusedBatchNorm2d.
 """
 
 import torch
 import torch.nn as nn
 
-from chainermn_amd.functions.batch_normalization import multi_node_batch_norm
+from chainermn_amd.links.fused_batch_norm import FusedBatchNorm2d
 
 
-class MultiNodeBatchNormalization(nn.Module):
+class MultiNodeBatchNormalization(FusedBatchNorm2d):
+
+    _running_unbias = 1.0  # reference keeps biased batch var
 
     def __init__(self, size, comm, decay=0.9, eps=2e-5, dtype=None,
                  use_gamma=True, use_beta=True, communication_backend='auto'):
-        super().__init__()
         if communication_backend not in ('auto', 'rccl', 'nccl', 'gloo'):
             raise ValueError(
                 'communication_backend must be auto/rccl/nccl/gloo')
+        super().__init__(size, eps=eps, momentum=1.0 - decay,
+                         affine=True, track_running_stats=True)
         self.comm = comm
-        self.eps = eps
-        self.decay = decay  # torch momentum = 1 - decay
-        self.num_features = size
+        self.decay = decay
         dtype = dtype or torch.float32
         if use_gamma:
             self.weight = nn.Parameter(torch.ones(size, dtype=dtype))
@@ -34,40 +42,13 @@ class MultiNodeBatchNormalization(nn.Module):
             self.bias = nn.Parameter(torch.zeros(size, dtype=dtype))
         else:
             self.register_parameter('bias', None)
-        self.register_buffer('running_mean', torch.zeros(size))
-        self.register_buffer('running_var', torch.ones(size))
-        self.register_buffer('num_batches_tracked',
-                             torch.tensor(0, dtype=torch.long))
-
-    def forward(self, x):
-        if self.training:
-            gamma = (self.weight if self.weight is not None
-                     else torch.ones(self.num_features, device=x.device,
-                                     dtype=x.dtype))
-            beta = (self.bias if self.bias is not None
-                    else torch.zeros(self.num_features, device=x.device,
-                                     dtype=x.dtype))
-            y, mean, var = multi_node_batch_norm(x, gamma, beta, self.comm,
-                                                 self.eps)
-            with torch.no_grad():
-                m = 1.0 - self.decay
-                self.running_mean.mul_(self.decay).add_(mean.detach(),
-                                                        alpha=m)
-                self.running_var.mul_(self.decay).add_(var.detach(), alpha=m)
-                self.num_batches_tracked += 1
-            return y
-        else:
-            shape = [1] * x.dim()
-            shape[1] = self.num_features
-            rstd = torch.rsqrt(self.running_var + self.eps)
-            y = (x.float() - self.running_mean.view(shape)) \
-                * rstd.view(shape)
-            if self.weight is not None:
-                y = y * self.weight.float().view(shape)
-            if self.bias is not None:
-                y = y + self.bias.float().view(shape)
-            return y.to(x.dtype)
 
     def extra_repr(self):
         return '{}, decay={}, eps={}, comm_size={}'.format(
             self.num_features, self.decay, self.eps, self.comm.size)
+
+
+class MultiNodeBNReLU2d(MultiNodeBatchNormalization):
+    """y = relu(global-batch-bn(x)) in one kernel pass."""
+
+    _relu_mode = 1

@@ -37,26 +37,40 @@ def _empty(x):
 
 
 class _FusedBNAct(torch.autograd.Function):
-    """relu_mode: 0 = bn, 1 = bn+relu (affine mask), 2 = bn+add+relu."""
+    """relu_mode: 0 = bn, 1 = bn+relu (affine mask), 2 = bn+add+relu.
+
+    ``comm`` non-None → multi-node BN: the (Σx, Σx²) buffer is summed
+    across ranks before finalize, and backward all-reduces (Σgy', Σgy'x̂)
+    — exact global-batch statistics (assumes rank-equal local batch, as
+    the reference did; use scatter_dataset(force_equal_length=True)).
+    ``unbias``: running_var correction (n/(n-1) torch-style, or 1.0 for
+    the reference's biased convention).
+    """
 
     @staticmethod
     def forward(ctx, x, res, gamma, beta, eps, relu_mode, momentum,
-                r_mean, r_var):
+                r_mean, r_var, comm=None, unbias=None):
         C = x.shape[1]
         count = x.numel() // C
+        world = comm.size if comm is not None else 1
+        total = count * world
         gamma_f = gamma.to(x.dtype)
         beta_f = beta.to(x.dtype)
+        if unbias is None:
+            unbias = total / max(total - 1, 1)
         if x.is_cuda:
             ext = ops.require()
             buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
             ext.bn_stats_into(x, buf)
+            if world > 1:
+                import torch.distributed as dist
+                dist.all_reduce(buf, group=comm.group)
             if r_mean is not None:
-                unbias = count / max(count - 1, 1)
                 mean, rstd, var = ext.bn_finalize(
-                    buf, float(count), eps, momentum, unbias, r_mean, r_var)
+                    buf, float(total), eps, momentum, unbias, r_mean, r_var)
             else:
                 mean, rstd, var = ext.bn_finalize(
-                    buf, float(count), eps, 0.0, 1.0, _empty(buf),
+                    buf, float(total), eps, 0.0, 1.0, _empty(buf),
                     _empty(buf))
             y = ext.bn_fwd_act(x, mean, rstd, gamma_f, beta_f,
                                res if res is not None else _empty(x),
@@ -64,8 +78,15 @@ class _FusedBNAct(torch.autograd.Function):
         else:
             dims = [d for d in range(x.dim()) if d != 1]
             xf = x.float()
-            mean = xf.mean(dim=dims)
-            var = xf.var(dim=dims, unbiased=False)
+            s_sum = xf.sum(dim=dims)
+            sq_sum = (xf * xf).sum(dim=dims)
+            if world > 1:
+                import torch.distributed as dist
+                buf = torch.cat([s_sum, sq_sum])
+                dist.all_reduce(buf, group=comm.group)
+                s_sum, sq_sum = buf[:C], buf[C:]
+            mean = s_sum / total
+            var = (sq_sum / total - mean * mean).clamp_min(0)
             rstd = torch.rsqrt(var + eps)
             y = (xf - _shape(mean, x)) * _shape(rstd, x) \
                 * _shape(gamma_f.float(), x) + _shape(beta_f.float(), x)
@@ -76,7 +97,6 @@ class _FusedBNAct(torch.autograd.Function):
             y = y.to(x.dtype)
             if r_mean is not None:
                 with torch.no_grad():
-                    unbias = count / max(count - 1, 1)
                     r_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
                     r_var.mul_(1 - momentum).add_(var * unbias,
                                                   alpha=momentum)
@@ -85,7 +105,8 @@ class _FusedBNAct(torch.autograd.Function):
         else:
             ctx.save_for_backward(x, gamma, beta, mean, rstd)
         ctx.relu_mode = relu_mode
-        ctx.count = count
+        ctx.count = total
+        ctx.comm = comm if world > 1 else None
         ctx.has_res = res is not None
         return y
 
@@ -107,6 +128,9 @@ class _FusedBNAct(torch.autograd.Function):
             ext.bn_bwd_reduce_act_into(
                 x, gy, mean, rstd, gamma_f, beta_f,
                 yout if yout is not None else _empty(x), relu_mode, buf)
+            if ctx.comm is not None:
+                import torch.distributed as dist
+                dist.all_reduce(buf, group=ctx.comm.group)
             gbeta = buf[:C]
             ggamma = buf[C:]
             gx, gres = ext.bn_bwd_input_act(
@@ -125,13 +149,18 @@ class _FusedBNAct(torch.autograd.Function):
                 gyf = gyf * (yout.float() > 0)
             gbeta = gyf.sum(dim=dims)
             ggamma = (gyf * xhat).sum(dim=dims)
+            if ctx.comm is not None:
+                import torch.distributed as dist
+                buf = torch.cat([gbeta, ggamma])
+                dist.all_reduce(buf, group=ctx.comm.group)
+                gbeta, ggamma = buf[:C], buf[C:]
             gx = (_shape(gamma_f.float() * rstd, x)
                   * (gyf - (xhat * _shape(ggamma, x)
                             + _shape(gbeta, x)) / count)).to(x.dtype)
             gres = gyf.to(x.dtype) if ctx.has_res else None
         g_res = gres if ctx.has_res else None
         return (gx, g_res, ggamma.to(gamma.dtype), gbeta.to(beta.dtype),
-                None, None, None, None, None)
+                None, None, None, None, None, None, None)
 
 
 class FusedBatchNorm2d(nn.Module):
@@ -139,6 +168,8 @@ class FusedBatchNorm2d(nn.Module):
     unbiased running_var) on the fused gfx950 kernels."""
 
     _relu_mode = 0
+    comm = None            # set → global-batch (multi-node) statistics
+    _running_unbias = None  # None = torch n/(n-1); 1.0 = biased (chainermn)
 
     def __init__(self, num_features, eps=1e-5, momentum=0.1, affine=True,
                  track_running_stats=True):
@@ -194,7 +225,8 @@ class FusedBatchNorm2d(nn.Module):
         gamma, beta = self._gamma_beta(x)
         r_mean, r_var, m = self._running()
         y = _FusedBNAct.apply(x, res, gamma, beta, self.eps, relu_mode, m,
-                              r_mean, r_var)
+                              r_mean, r_var, self.comm,
+                              self._running_unbias)
         if self.track_running_stats:
             self.num_batches_tracked += 1
         return y

@@ -17,14 +17,12 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, in_ch, width, stride=1, downsample=None,
-                 norm_layer=nn.BatchNorm2d, fused_ops=False):
+                 norm_layer=nn.BatchNorm2d, fused_ops=False, comm=None):
         super().__init__()
         out_ch = width * self.expansion
         self.fused_ops = fused_ops
         if fused_ops:
-            from chainermn_amd.links.fused_batch_norm import (
-                FusedBatchNorm2d, FusedBNReLU2d)
-            bn_relu, bn_plain = FusedBNReLU2d, FusedBatchNorm2d
+            bn_relu, bn_plain = _fused_bn_classes(comm)
         else:
             bn_relu = bn_plain = norm_layer
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
@@ -56,16 +54,27 @@ class Bottleneck(nn.Module):
         return self.relu(out)
 
 
+def _fused_bn_classes(comm):
+    """(bn_relu, bn_plain) classes: local fused BN, or multi-node fused BN
+    when a communicator is given (global-batch statistics over RCCL)."""
+    if comm is not None and comm.size > 1:
+        from chainermn_amd.links.batch_normalization import (
+            MultiNodeBatchNormalization, MultiNodeBNReLU2d)
+        return (lambda c: MultiNodeBNReLU2d(c, comm),
+                lambda c: MultiNodeBatchNormalization(c, comm))
+    from chainermn_amd.links.fused_batch_norm import (FusedBatchNorm2d,
+                                                      FusedBNReLU2d)
+    return FusedBNReLU2d, FusedBatchNorm2d
+
+
 class ResNet50(nn.Module):
 
     def __init__(self, num_classes=1000, norm_layer=nn.BatchNorm2d,
-                 zero_init_residual=True, fused_ops=False):
+                 zero_init_residual=True, fused_ops=False, comm=None):
         super().__init__()
+        self._comm = comm
         if fused_ops:
-            from chainermn_amd.links.fused_batch_norm import (
-                FusedBatchNorm2d, FusedBNReLU2d)
-            self._stem_bn = FusedBNReLU2d
-            norm_layer = FusedBatchNorm2d
+            self._stem_bn, norm_layer = _fused_bn_classes(comm)
         else:
             self._stem_bn = norm_layer
         self._norm = norm_layer
@@ -103,11 +112,13 @@ class ResNet50(nn.Module):
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
                 norm_layer(out_ch))
         layers = [Bottleneck(self.in_ch, width, stride, downsample,
-                             norm_layer, fused_ops=self._fused_ops)]
+                             norm_layer, fused_ops=self._fused_ops,
+                             comm=self._comm)]
         self.in_ch = out_ch
         for _ in range(1, blocks):
             layers.append(Bottleneck(out_ch, width, norm_layer=norm_layer,
-                                     fused_ops=self._fused_ops))
+                                     fused_ops=self._fused_ops,
+                                     comm=self._comm))
         return nn.Sequential(*layers)
 
     def forward(self, x):

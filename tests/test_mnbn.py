@@ -117,3 +117,38 @@ def _create_mnbn_worker(rank, size):
 
 def test_create_mnbn_model():
     run_distributed(_create_mnbn_worker, 2)
+
+
+def _fused_mnbn_worker(rank, size, x_all_np, gy_all_np):
+    """Fused-path MNBN relu variant across 2 ranks vs composed serial."""
+    from chainermn_amd.links.batch_normalization import MultiNodeBNReLU2d
+    comm = chainermn_amd.create_communicator('naive')
+    x_all = torch.from_numpy(x_all_np)
+    gy_all = torch.from_numpy(gy_all_np)
+    x = x_all[rank * B:(rank + 1) * B].clone().requires_grad_(True)
+    gy = gy_all[rank * B:(rank + 1) * B]
+    bn = MultiNodeBNReLU2d(C, comm)
+    y = bn(x)
+    y.backward(gy)
+    return {'y': y.detach(), 'gx': x.grad, 'ggamma': bn.weight.grad}
+
+
+def test_fused_mnbn_relu_matches_global_batch():
+    size = 2
+    rng = numpy.random.RandomState(3)
+    x_all = rng.randn(size * B, C, H, W).astype(numpy.float32)
+    gy_all = rng.randn(size * B, C, H, W).astype(numpy.float32)
+    results = run_distributed(_fused_mnbn_worker, size, args=(x_all, gy_all))
+    x = torch.from_numpy(x_all).clone().requires_grad_(True)
+    bn = nn.BatchNorm2d(C, eps=2e-5)
+    y = torch.relu(bn(x))
+    y.backward(torch.from_numpy(gy_all))
+    for rank in range(size):
+        r = results[rank]
+        assert torch.allclose(torch.from_numpy(r['y']),
+                              y.detach()[rank * B:(rank + 1) * B],
+                              atol=1e-5)
+        assert torch.allclose(torch.from_numpy(r['gx']),
+                              x.grad[rank * B:(rank + 1) * B], atol=1e-5)
+        assert torch.allclose(torch.from_numpy(r['ggamma']),
+                              bn.weight.grad, atol=1e-4)
