@@ -51,6 +51,9 @@ std::tuple<torch::Tensor, torch::Tensor> bn_bwd_reduce(torch::Tensor x,
                                                        torch::Tensor rstd);
 torch::Tensor bn_fwd(torch::Tensor x, torch::Tensor mean, torch::Tensor rstd,
                      torch::Tensor gamma, torch::Tensor beta);
+std::tuple<torch::Tensor, torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x);
+torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
+                               long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
                            torch::Tensor mean, torch::Tensor rstd,
                            torch::Tensor gamma, torch::Tensor gbeta,
@@ -90,4 +93,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "y = act(bn(x) [+ res]) (fused relu / residual add)");
   m.def("bn_bwd_input_act", &bn_bwd_input_act,
         "BN input grad with fused relu mask [+ gres]");
+  m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd,
+        "channels-last 3x3/s2/p1 maxpool forward (+window codes)");
+  m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd,
+        "gather-based maxpool backward (no atomics)");
 }

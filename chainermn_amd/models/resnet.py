@@ -84,7 +84,11 @@ class ResNet50(nn.Module):
         self.bn1 = self._stem_bn(64)
         self.relu = (nn.Identity() if fused_ops
                      else nn.ReLU(inplace=True))
-        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        if fused_ops:
+            from chainermn_amd.links.fused_pool import FusedMaxPool2d
+            self.maxpool = FusedMaxPool2d(3, stride=2, padding=1)
+        else:
+            self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, 3, stride=1)
         self.layer2 = self._make_layer(128, 4, stride=2)
         self.layer3 = self._make_layer(256, 6, stride=2)

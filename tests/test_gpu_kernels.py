@@ -540,3 +540,26 @@ def test_bucketed_overlap_gpu_single_rank():
     torch.cuda.synchronize()
     for p, r in zip(model.parameters(), ref.parameters()):
         assert torch.allclose(p.data, r.data, atol=1e-5)
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_maxpool_vs_torch(dtype):
+    from chainermn_amd.links import FusedMaxPool2d
+    torch.manual_seed(41)
+    for N, C, H, W in [(4, 64, 112, 112), (2, 8, 9, 9), (1, 16, 7, 10)]:
+        x32 = torch.randn(N, C, H, W, device=DEV)
+        x = x32.to(dtype).to(memory_format=torch.channels_last)
+        x.requires_grad_(True)
+        xr = x32.to(dtype).clone().requires_grad_(True)
+        pool = FusedMaxPool2d(3, 2, 1)
+        y = pool(x)
+        yr = torch.nn.functional.max_pool2d(xr, 3, 2, 1)
+        assert y.shape == yr.shape
+        assert torch.equal(y.float(),
+                           yr.float().to(memory_format=torch.channels_last)
+                           .float()), (N, C, H, W)
+        gy = torch.randn_like(yr)
+        y.backward(gy.to(memory_format=torch.channels_last))
+        yr.backward(gy)
+        assert torch.allclose(x.grad.float(), xr.grad.float(), atol=1e-3), \
+            (N, C, H, W, (x.grad.float() - xr.grad.float()).abs().max())
