@@ -1,0 +1,70 @@
+"""FusedConv1x1 — 1x1 convolutions as hipBLASLt GEMMs.
+
+A channels-last 1x1/stride-1 convolution IS a GEMM over
+[N*H*W, Cin] x [Cin, Cout] (the permuted view of a channels_last tensor is
+free).  Measured on MI355X (batch 512, see profiles/): hipBLASLt GEMMs run
+1.4-1.6x faster than MIOpen's igemm conv kernels for ResNet-50's 1x1
+shapes on forward and data-grad.  The weight-grad GEMM shape
+([Cout, M] x [M, Cin], huge K, tiny output) is pathological for a plain
+GEMM call, so backward routes grad_weight through MIOpen's tuned
+split-K wrw kernels via aten::convolution_backward.
+"""
+
+import torch
+import torch.nn as nn
+
+
+class _Conv1x1Fn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, x, w):
+        # x: channels_last [N, C, H, W]; its (0,2,3,1) permute is a free
+        # contiguous view
+        N, C, H, W = x.shape
+        x2 = x.permute(0, 2, 3, 1).reshape(-1, C)
+        w2 = w.reshape(w.shape[0], C)
+        y2 = x2 @ w2.t()
+        y = y2.view(N, H, W, -1).permute(0, 3, 1, 2)
+        ctx.save_for_backward(x, w)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        N, C, H, W = x.shape
+        Cout = w.shape[0]
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        gy2 = gy.permute(0, 2, 3, 1).reshape(-1, Cout)
+        w2 = w.reshape(Cout, C)
+        gx2 = gy2 @ w2
+        gx = gx2.view(N, H, W, C).permute(0, 3, 1, 2)
+        # weight grad: MIOpen's split-K wrw kernels (the plain GEMM form
+        # [Cout, M] x [M, Cin] with M ~ 1e6 is ~10x slower)
+        _, gw, _ = torch.ops.aten.convolution_backward(
+            gy, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            [False, True, False])
+        return gx, gw
+
+
+class FusedConv1x1(nn.Module):
+    """Drop-in for nn.Conv2d(cin, cout, 1, bias=False) on channels_last."""
+
+    def __init__(self, in_channels, out_channels):
+        super().__init__()
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        # identical initialization (and RNG consumption) to
+        # nn.Conv2d(cin, cout, 1, bias=False)
+        ref = nn.Conv2d(in_channels, out_channels, 1, bias=False)
+        self.weight = ref.weight
+
+    def forward(self, x):
+        if x.is_cuda and x.dim() == 4 \
+                and x.is_contiguous(memory_format=torch.channels_last) \
+                and not x.is_contiguous():
+            return _Conv1x1Fn.apply(x, self.weight)
+        return torch.nn.functional.conv2d(x, self.weight)
+
+    def extra_repr(self):
+        return '{}, {}, kernel_size=1 (GEMM)'.format(self.in_channels,
+                                                     self.out_channels)

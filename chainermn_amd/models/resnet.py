@@ -23,14 +23,17 @@ class Bottleneck(nn.Module):
         self.fused_ops = fused_ops
         if fused_ops:
             bn_relu, bn_plain = _fused_bn_classes(comm)
+            from chainermn_amd.links.fused_conv import FusedConv1x1
+            conv1x1 = FusedConv1x1
         else:
             bn_relu = bn_plain = norm_layer
-        self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
+            conv1x1 = (lambda ci, co: nn.Conv2d(ci, co, 1, bias=False))
+        self.conv1 = conv1x1(in_ch, width)
         self.bn1 = bn_relu(width)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
         self.bn2 = bn_relu(width)
-        self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
+        self.conv3 = conv1x1(width, out_ch)
         self.bn3 = bn_plain(out_ch)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
@@ -95,8 +98,9 @@ class ResNet50(nn.Module):
         self.layer4 = self._make_layer(512, 3, stride=2)
         self.avgpool = nn.AdaptiveAvgPool2d(1)
         self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
+        from chainermn_amd.links.fused_conv import FusedConv1x1
         for m in self.modules():
-            if isinstance(m, nn.Conv2d):
+            if isinstance(m, (nn.Conv2d, FusedConv1x1)):
                 nn.init.kaiming_normal_(m.weight, mode='fan_out',
                                         nonlinearity='relu')
             elif isinstance(m, nn.BatchNorm2d):

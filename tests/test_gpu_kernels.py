@@ -563,3 +563,27 @@ def test_fused_maxpool_vs_torch(dtype):
         yr.backward(gy)
         assert torch.allclose(x.grad.float(), xr.grad.float(), atol=1e-3), \
             (N, C, H, W, (x.grad.float() - xr.grad.float()).abs().max())
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_conv1x1_vs_conv2d(dtype):
+    from chainermn_amd.links import FusedConv1x1
+    torch.manual_seed(51)
+    N, C, H, W, Cout = 8, 64, 14, 14, 128
+    x32 = torch.randn(N, C, H, W, device=DEV)
+    gy32 = torch.randn(N, Cout, H, W, device=DEV)
+    conv = FusedConv1x1(C, Cout).to(DEV, dtype)
+    x = x32.to(dtype).to(memory_format=torch.channels_last)
+    x.requires_grad_(True)
+    xr = x32.to(dtype).clone().requires_grad_(True)
+    y = conv(x)
+    yr = torch.nn.functional.conv2d(xr, conv.weight)
+    tol = 5e-2 if dtype == torch.bfloat16 else 1e-4
+    assert torch.allclose(y.float(), yr.float(), atol=tol, rtol=1e-2)
+    w_grad_ref = torch.autograd.grad(
+        yr, [xr, conv.weight], gy32.to(dtype), retain_graph=False)
+    y.backward(gy32.to(dtype).to(memory_format=torch.channels_last))
+    assert torch.allclose(x.grad.float(), w_grad_ref[0].float(), atol=tol,
+                          rtol=1e-2)
+    assert torch.allclose(conv.weight.grad.float(),
+                          w_grad_ref[1].float(), atol=tol * 10, rtol=5e-2)
