@@ -20,6 +20,10 @@ def scatter_dataset(dataset, comm, root=0, shuffle=False, seed=None,
     ``force_equal_length`` shards wrap-pad to identical length so every
     rank runs the same number of iterations — keeping per-iteration
     collective counts rank-uniform).
+
+    Note: items are pickled.  A torch tensor VIEW pickles its entire
+    backing storage — materialize items (``.clone()``) or use
+    ``scatter_index`` for large root-side datasets.
     """
     if comm.size == 1:
         return dataset

@@ -33,7 +33,9 @@ def synthetic_mnist(n=4096, seed=0):
     g = torch.Generator().manual_seed(seed)
     xs = torch.rand(n, 784, generator=g)
     ys = torch.arange(n) % 10
-    return [(xs[i], int(ys[i])) for i in range(n)]
+    # clone(): pickling a VIEW serializes the whole backing storage
+    # (12 MB per item here) — clone keeps shards compact
+    return [(xs[i].clone(), int(ys[i])) for i in range(n)]
 
 
 def main():
