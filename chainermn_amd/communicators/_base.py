@@ -10,7 +10,6 @@ shapes, dtypes sent ahead of payloads) is kept so send/recv carry arbitrary
 tensor tuples with no prior shape agreement.
 """
 
-import io
 import pickle
 
 import torch

@@ -6,7 +6,7 @@ Used by tests and single-GPU debugging without a launcher.
 
 import torch
 
-from chainermn_amd.communicators._base import CommunicatorBase, _MessageType
+from chainermn_amd.communicators._base import CommunicatorBase
 
 
 class DummyCommunicator(CommunicatorBase):

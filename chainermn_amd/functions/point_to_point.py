@@ -14,8 +14,6 @@ reverse-topological replay preserves.
 
 import torch
 
-from chainermn_amd.functions.pseudo_connect import pseudo_connect
-
 
 class _Send(torch.autograd.Function):
 

@@ -12,7 +12,6 @@ The backward graph replays sends/recvs in reverse delegate order, which is
 what makes the RCCL p2p (tag-less, issue-ordered) deadlock-free.
 """
 
-import torch
 import torch.nn as nn
 
 from chainermn_amd import functions

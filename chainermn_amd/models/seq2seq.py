@@ -9,7 +9,6 @@ returns (encoder_part, decoder_part) wrappers for a 2-rank split where the
 final hidden state crosses the xGMI link via differentiable send/recv.
 """
 
-import torch
 import torch.nn as nn
 
 from chainermn_amd import functions
