@@ -38,20 +38,24 @@ class _Conv1x1Fn(torch.autograd.Function):
         w2 = w.reshape(Cout, C)
         gx2 = gy2 @ w2
         gx = gx2.view(N, H, W, C).permute(0, 3, 1, 2)
-        # weight grad: the plain GEMM form ([Cout, M] x [M, Cin], M ~ 1e6,
-        # tiny output) is ~10x slower than split-K.  A batched-GEMM
-        # split-K (fp32 partial sum) beats MIOpen's wrw kernels by
-        # 8-30% on the ResNet shapes (see profiles/).
-        M = gy2.shape[0]
-        x2 = x.permute(0, 2, 3, 1).reshape(M, C)
-        S = next((s for s in (128, 64, 32, 16, 8, 4, 2, 1) if M % s == 0))
-        if S > 1:
+        # weight grad: MIOpen's tuned split-K wrw kernels.  (The plain
+        # GEMM form [Cout, M] x [M, Cin] with M ~ 1e6 is ~10x slower; a
+        # hipBLASLt bmm split-K looked 8-30% faster in isolation but
+        # measured 2 ms/step SLOWER end to end — within-box A/B in
+        # profiles/README.md — so MIOpen stays.)
+        import os
+        if os.environ.get('CHAINERMN_AMD_WGRAD') == 'bmm':
+            M = gy2.shape[0]
+            x2 = x.permute(0, 2, 3, 1).reshape(M, C)
+            S = next((s for s in (128, 64, 32, 16, 8, 4, 2, 1)
+                      if M % s == 0))
             part = torch.bmm(gy2.view(S, M // S, Cout).transpose(1, 2),
                              x2.view(S, M // S, C))
-            gw2 = part.float().sum(0)
-        else:
-            gw2 = (gy2.t().float() @ x2.float())
-        gw = gw2.to(w.dtype).view_as(w)
+            gw = part.float().sum(0).to(w.dtype).view_as(w)
+            return gx, gw
+        _, gw, _ = torch.ops.aten.convolution_backward(
+            gy, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            [False, True, False])
         return gx, gw
 
 
