@@ -111,6 +111,23 @@ __global__ void fused_adam_kernel(float* __restrict__ p,
   }
 }
 
+__device__ __forceinline__ void load4s(const float* p, float (&r)[4]) {
+  const float4 v = *reinterpret_cast<const float4*>(p);
+  r[0] = v.x; r[1] = v.y; r[2] = v.z; r[3] = v.w;
+}
+template <typename T16>
+__device__ __forceinline__ void load4s(const T16* p, float (&r)[4]) {
+  const uint2 v = *reinterpret_cast<const uint2*>(p);
+  const unsigned u[2] = {v.x, v.y};
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    r[2 * h] = float(T16(uint16_t(u[h] & 0xffffu),
+                         typename T16::from_bits_t()));
+    r[2 * h + 1] = float(T16(uint16_t(u[h] >> 16),
+                             typename T16::from_bits_t()));
+  }
+}
+
 // Per-segment sum of squares over a flat buffer.  Each block owns a
 // contiguous chunk; for every segment overlapping the chunk it
 // block-reduces the local sum and atomically adds to out[seg].
@@ -130,7 +147,15 @@ __global__ void segment_sq_norms_kernel(const T* __restrict__ flat,
     const long s0 = max(c0, offsets[seg]);
     const long s1 = min(c1, offsets[seg] + sizes[seg]);
     float acc = 0.f;
-    for (long i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
+    // vectorized main body (s0 is 4-aligned: chunk and slot starts are
+    // multiples of 256/ALIGN elements), scalar tail
+    const long n4 = s0 + ((s1 - s0) / 4) * 4;
+    for (long i = s0 + threadIdx.x * 4; i < n4; i += blockDim.x * 4) {
+      float r[4];
+      load4s(flat + i, r);
+      acc += r[0] * r[0] + r[1] * r[1] + r[2] * r[2] + r[3] * r[3];
+    }
+    for (long i = n4 + threadIdx.x; i < s1; i += blockDim.x) {
       const float x = float(flat[i]);
       acc += x * x;
     }

@@ -22,60 +22,60 @@
 
 namespace {
 
-template <typename SRC, typename DST>
-__global__ void batched_pack_kernel(const long* __restrict__ ptrs,
+// Each block walks a CONTIGUOUS chunk of the flat space (coalesced: the
+// 256 lanes cover 2048 consecutive elements per iteration) and caches the
+// segment bounds — one binary search per segment crossing instead of per
+// vector (the grid-stride variant measured 46% of HBM peak; per-iteration
+// search + scattered chunks were the cost).
+template <typename SRC, typename DST, bool PACK>
+__global__ void batched_copy_kernel(const long* __restrict__ ptrs,
                                     const long* __restrict__ sizes,
                                     const long* __restrict__ offsets,
                                     int n_params,
-                                    DST* __restrict__ out,
-                                    long total8) {
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long v = blockIdx.x * (long)blockDim.x + threadIdx.x; v < total8;
-       v += stride) {
+                                    DST* __restrict__ flat_dst,
+                                    const SRC* __restrict__ flat_src,
+                                    float scale,
+                                    long chunk8, long total8) {
+  const long v0 = blockIdx.x * chunk8;
+  const long v1 = min(v0 + chunk8, total8);
+  int seg = -1;
+  long seg_start = 0, seg_end = -1, size = 0;
+  const SRC* __restrict__ psrc = nullptr;
+  DST* __restrict__ pdst = nullptr;
+  for (long v = v0 + threadIdx.x; v < v1; v += blockDim.x) {
     const long base = v * 8;
-    const int seg = find_segment(offsets, n_params, base);
-    const long local = base - offsets[seg];
-    const long size = sizes[seg];
-    const SRC* __restrict__ src = reinterpret_cast<const SRC*>(ptrs[seg]);
-    float r[8];
-    if (local + 8 <= size) {
-      V8<SRC>::load(src + local, r);
-    } else {
-#pragma unroll
-      for (int k = 0; k < 8; ++k)
-        r[k] = (local + k < size) ? float(src[local + k]) : 0.f;
+    if (base < seg_start || base >= seg_end) {
+      seg = find_segment(offsets, n_params, base);
+      seg_start = offsets[seg];
+      seg_end = (seg + 1 < n_params) ? offsets[seg + 1]
+                                     : (total8 * 8);
+      size = sizes[seg];
+      if (PACK) psrc = reinterpret_cast<const SRC*>(ptrs[seg]);
+      else pdst = reinterpret_cast<DST*>(ptrs[seg]);
     }
-    V8<DST>::store(out + base, r);
-  }
-}
-
-template <typename SRC, typename DST>
-__global__ void batched_unpack_kernel(const long* __restrict__ ptrs,
-                                      const long* __restrict__ sizes,
-                                      const long* __restrict__ offsets,
-                                      int n_params,
-                                      const SRC* __restrict__ in,
-                                      float scale,
-                                      long total8) {
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long v = blockIdx.x * (long)blockDim.x + threadIdx.x; v < total8;
-       v += stride) {
-    const long base = v * 8;
-    const int seg = find_segment(offsets, n_params, base);
-    const long local = base - offsets[seg];
-    const long size = sizes[seg];
-    if (local >= size) continue;  // pad region
-    DST* __restrict__ dst = reinterpret_cast<DST*>(ptrs[seg]);
+    const long local = base - seg_start;
     float r[8];
-    V8<SRC>::load(in + base, r);
+    if (PACK) {
+      if (local + 8 <= size) {
+        V8<SRC>::load(psrc + local, r);
+      } else {
 #pragma unroll
-    for (int k = 0; k < 8; ++k) r[k] *= scale;
-    if (local + 8 <= size) {
-      V8<DST>::store(dst + local, r);
+        for (int k = 0; k < 8; ++k)
+          r[k] = (local + k < size) ? float(psrc[local + k]) : 0.f;
+      }
+      V8<DST>::store(flat_dst + base, r);
     } else {
+      if (local >= size) continue;  // pad region
+      V8<SRC>::load(flat_src + base, r);
 #pragma unroll
-      for (int k = 0; k < 8; ++k)
-        if (local + k < size) dst[local + k] = DST(r[k]);
+      for (int k = 0; k < 8; ++k) r[k] *= scale;
+      if (local + 8 <= size) {
+        V8<DST>::store(pdst + local, r);
+      } else {
+#pragma unroll
+        for (int k = 0; k < 8; ++k)
+          if (local + k < size) pdst[local + k] = DST(r[k]);
+      }
     }
   }
 }
@@ -117,12 +117,15 @@ void batched_pack(torch::Tensor ptrs, torch::Tensor sizes,
   const int n_params = (int)ptrs.numel();
   const long total8 = out.numel() / 8;
   auto stream = c10::hip::getCurrentHIPStream();
+  const long chunk8 = std::max((total8 + 2047) / 2048, 32L);
+  const int grid = (int)((total8 + chunk8 - 1) / chunk8);
   DISPATCH_PAIR(src_dtype, out.scalar_type(), [&] {
-    hipLaunchKernelGGL((batched_pack_kernel<src_t, dst_t>),
-                       dim3(cmn_grid(total8, CMN_BLOCK)), dim3(CMN_BLOCK), 0,
+    hipLaunchKernelGGL((batched_copy_kernel<src_t, dst_t, true>),
+                       dim3(grid), dim3(CMN_BLOCK), 0,
                        stream.stream(), ptrs.data_ptr<long>(),
                        sizes.data_ptr<long>(), offsets.data_ptr<long>(),
-                       n_params, (dst_t*)out.data_ptr(), total8);
+                       n_params, (dst_t*)out.data_ptr(), nullptr, 1.f,
+                       chunk8, total8);
   });
 }
 
@@ -133,12 +136,15 @@ void batched_unpack(torch::Tensor ptrs, torch::Tensor sizes,
   const int n_params = (int)ptrs.numel();
   const long total8 = in.numel() / 8;
   auto stream = c10::hip::getCurrentHIPStream();
+  const long chunk8 = std::max((total8 + 2047) / 2048, 32L);
+  const int grid = (int)((total8 + chunk8 - 1) / chunk8);
   DISPATCH_PAIR(in.scalar_type(), dst_dtype, [&] {
-    hipLaunchKernelGGL((batched_unpack_kernel<src_t, dst_t>),
-                       dim3(cmn_grid(total8, CMN_BLOCK)), dim3(CMN_BLOCK), 0,
+    hipLaunchKernelGGL((batched_copy_kernel<src_t, dst_t, false>),
+                       dim3(grid), dim3(CMN_BLOCK), 0,
                        stream.stream(), ptrs.data_ptr<long>(),
                        sizes.data_ptr<long>(), offsets.data_ptr<long>(),
-                       n_params, (const src_t*)in.data_ptr(), (float)scale,
+                       n_params, (dst_t*)in.data_ptr() /*unused*/,
+                       (const src_t*)in.data_ptr(), (float)scale, chunk8,
                        total8);
   });
 }
