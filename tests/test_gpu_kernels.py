@@ -587,3 +587,21 @@ def test_fused_conv1x1_vs_conv2d(dtype):
                           rtol=1e-2)
     assert torch.allclose(conv.weight.grad.float(),
                           w_grad_ref[1].float(), atol=tol * 10, rtol=5e-2)
+
+
+def test_flat_and_staged_communicators_world1():
+    """flat/single_node/hierarchical/two_dimensional mean_grad on a
+    1-rank GPU world (exercises the pack->reduce->unpack machinery and
+    sub-group construction with RCCL available)."""
+    import chainermn_amd
+    import torch.nn as nn
+    for name in ['flat', 'single_node', 'hierarchical', 'two_dimensional',
+                 'non_cuda_aware', 'pure_rccl']:
+        comm = chainermn_amd.create_communicator(name)
+        model = nn.Linear(17, 5).to(DEV)
+        for p in model.parameters():
+            p.grad = torch.full_like(p, 2.0)
+        comm.multi_node_mean_grad(model)
+        for p in model.parameters():
+            assert torch.allclose(p.grad, torch.full_like(p.grad, 2.0),
+                                  atol=1e-2), name
