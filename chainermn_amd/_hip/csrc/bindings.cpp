@@ -52,6 +52,7 @@ std::tuple<torch::Tensor, torch::Tensor> bn_bwd_reduce(torch::Tensor x,
 torch::Tensor bn_fwd(torch::Tensor x, torch::Tensor mean, torch::Tensor rstd,
                      torch::Tensor gamma, torch::Tensor beta);
 std::tuple<torch::Tensor, torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x);
+torch::Tensor gemm_bt_bf16(torch::Tensor a, torch::Tensor b);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -97,4 +98,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "channels-last 3x3/s2/p1 maxpool forward (+window codes)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd,
         "gather-based maxpool backward (no atomics)");
+  m.def("gemm_bt_bf16", &gemm_bt_bf16,
+        "hand-written MFMA bf16 GEMM: D = A @ B^T (128x128x64 tile, "
+        "glds staging, XOR-swizzled LDS, XCD-aware grid)");
 }

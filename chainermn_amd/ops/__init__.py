@@ -210,3 +210,14 @@ def lars_update(p, g, v, offsets, sizes, p_sqn, g_sqn, lr, momentum,
         ps.sub_(vs)
     if p_lowp is not None:
         p_lowp.copy_(p.to(p_lowp.dtype))
+
+
+def gemm_bt(a, b):
+    """D = a @ b.T via the hand-written MFMA kernel when shapes allow
+    (M%128 == N%128 == K%64 == 0, bf16, CUDA); hipBLASLt otherwise."""
+    if (a.is_cuda and a.dtype == torch.bfloat16
+            and a.shape[0] % 128 == 0 and b.shape[0] % 128 == 0
+            and a.shape[1] % 64 == 0 and a.is_contiguous()
+            and b.is_contiguous()):
+        return require().gemm_bt_bf16(a, b)
+    return a @ b.t()

@@ -605,3 +605,19 @@ def test_flat_and_staged_communicators_world1():
         for p in model.parameters():
             assert torch.allclose(p.grad, torch.full_like(p.grad, 2.0),
                                   atol=1e-2), name
+
+
+def test_mfma_gemm_bt_vs_matmul():
+    """Hand-written MFMA GEMM vs hipBLASLt, asymmetric operands
+    (transpose-detecting — guide §5.4 rule 16)."""
+    torch.manual_seed(61)
+    for M, N, K in [(128, 128, 64), (256, 384, 192), (512, 1024, 512)]:
+        a = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+        b = torch.randn(N, K, device=DEV, dtype=torch.bfloat16)
+        # asymmetric structure so a row/col swap cannot pass
+        a[0, :] += torch.arange(K, device=DEV, dtype=torch.bfloat16) * 0.01
+        d = ops.gemm_bt(a, b)
+        ref = a @ b.t()
+        err = (d.float() - ref.float()).abs().max()
+        ref_mag = ref.float().abs().max()
+        assert err <= 0.02 * ref_mag + 0.1, (M, N, K, err, ref_mag)

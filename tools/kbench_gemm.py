@@ -1,0 +1,20 @@
+"""MFMA GEMM microbench: hand-written kernel vs hipBLASLt."""
+import os, sys, time, torch
+sys.path.insert(0, os.environ.get("GRAFT_REPO_ROOT", "."))
+from chainermn_amd import ops
+
+def t(fn, n=30):
+    for _ in range(5): fn()
+    torch.cuda.synchronize(); t0=time.perf_counter()
+    for _ in range(n): fn()
+    torch.cuda.synchronize(); return (time.perf_counter()-t0)/n
+
+print("bf16 GEMM D = A @ B^T, uniform random [-1,1) operands:")
+for M, N, K in [(1024,1024,1024),(4096,4096,4096),(8192,8192,8192),
+                (401408,128,512),(100352,256,1024)]:
+    a = (torch.rand(M, K, device='cuda', dtype=torch.bfloat16)*2-1)
+    b = (torch.rand(N, K, device='cuda', dtype=torch.bfloat16)*2-1)
+    fl = 2.0*M*N*K
+    dt1 = t(lambda: ops.require().gemm_bt_bf16(a, b))
+    dt2 = t(lambda: a @ b.t())
+    print(f"  M{M} N{N} K{K}: ours {fl/dt1/1e12:7.1f} TF | hipBLASLt {fl/dt2/1e12:7.1f} TF")
