@@ -13,6 +13,8 @@ split-K wrw kernels via aten::convolution_backward.
 import torch
 import torch.nn as nn
 
+from chainermn_amd import ops
+
 
 class _Conv1x1Fn(torch.autograd.Function):
 
@@ -23,7 +25,7 @@ class _Conv1x1Fn(torch.autograd.Function):
         N, C, H, W = x.shape
         x2 = x.permute(0, 2, 3, 1).reshape(-1, C)
         w2 = w.reshape(w.shape[0], C)
-        y2 = x2 @ w2.t()
+        y2 = ops.gemm_bt(x2, w2)  # autotuned: MFMA kernel vs hipBLASLt
         y = y2.view(N, H, W, -1).permute(0, 3, 1, 2)
         ctx.save_for_backward(x, w)
         return y

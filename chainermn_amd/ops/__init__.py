@@ -212,12 +212,40 @@ def lars_update(p, g, v, offsets, sizes, p_sqn, g_sqn, lr, momentum,
         p_lowp.copy_(p.to(p_lowp.dtype))
 
 
+_GEMM_CHOICE = {}
+
+
+def _time_gpu(fn, n=5):
+    import time
+    fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(n):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / n
+
+
 def gemm_bt(a, b):
-    """D = a @ b.T via the hand-written MFMA kernel when shapes allow
-    (M%128 == N%128 == K%64 == 0, bf16, CUDA); hipBLASLt otherwise."""
-    if (a.is_cuda and a.dtype == torch.bfloat16
-            and a.shape[0] % 128 == 0 and b.shape[0] % 128 == 0
-            and a.shape[1] % 64 == 0 and a.is_contiguous()
-            and b.is_contiguous()):
+    """D = a @ b.T.  Eligible bf16 shapes (M%128 == N%128 == 0, K%64 == 0)
+    dispatch to whichever of {hand-written MFMA kernel, hipBLASLt} measured
+    faster for that (M, N, K) — decided once per shape at first call (the
+    MFMA kernel wins some conv-1x1 tall-skinny shapes by ~30%, hipBLASLt
+    wins large squares)."""
+    eligible = (a.is_cuda and a.dtype == torch.bfloat16
+                and a.shape[0] % 128 == 0 and b.shape[0] % 128 == 0
+                and a.shape[1] % 64 == 0 and a.is_contiguous()
+                and b.is_contiguous())
+    if not eligible:
+        return a @ b.t()
+    key = (a.shape[0], b.shape[0], a.shape[1])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        ext = require()
+        t_ours = _time_gpu(lambda: ext.gemm_bt_bf16(a, b))
+        t_blas = _time_gpu(lambda: a @ b.t())
+        choice = 'mfma' if t_ours < t_blas else 'blas'
+        _GEMM_CHOICE[key] = choice
+    if choice == 'mfma':
         return require().gemm_bt_bf16(a, b)
     return a @ b.t()
