@@ -38,7 +38,7 @@ class _Conv1x1Fn(torch.autograd.Function):
         gy = gy.contiguous(memory_format=torch.channels_last)
         gy2 = gy.permute(0, 2, 3, 1).reshape(-1, Cout)
         w2 = w.reshape(Cout, C)
-        gx2 = gy2 @ w2
+        gx2 = ops.gemm_nn(gy2, w2)  # autotuned
         gx = gx2.view(N, H, W, C).permute(0, 3, 1, 2)
         # weight grad: MIOpen's tuned split-K wrw kernels.  (The plain
         # GEMM form [Cout, M] x [M, Cin] with M ~ 1e6 is ~10x slower; a
