@@ -17,6 +17,11 @@ import torch.distributed as dist
 
 from chainermn_amd.utils import bootstrap
 
+# Object payloads travel in chunks of this many bytes (the reference's
+# chunked_bcast_obj kept single messages under MPI's INT32_MAX element
+# limit; gloo has the same practical bound).  Patchable for tests.
+OBJ_CHUNK_BYTES = 1 << 30
+
 _TORCH_DTYPES = {str(getattr(torch, n)): getattr(torch, n)
                  for n in dir(torch) if isinstance(getattr(torch, n), torch.dtype)}
 
@@ -124,14 +129,18 @@ class CommunicatorBase(object):
         length = torch.tensor([payload.numel()], dtype=torch.long)
         dst = self._global_rank(dest)
         dist.send(length, dst=dst, group=self._group, tag=tag)
-        dist.send(payload, dst=dst, group=self._group, tag=tag)
+        for off in range(0, max(payload.numel(), 1), OBJ_CHUNK_BYTES):
+            dist.send(payload[off:off + OBJ_CHUNK_BYTES], dst=dst,
+                      group=self._group, tag=tag)
 
     def recv_obj(self, source, tag=0):
         src = self._global_rank(source)
         length = torch.zeros(1, dtype=torch.long)
         dist.recv(length, src=src, group=self._group, tag=tag)
         payload = torch.empty(int(length.item()), dtype=torch.uint8)
-        dist.recv(payload, src=src, group=self._group, tag=tag)
+        for off in range(0, max(payload.numel(), 1), OBJ_CHUNK_BYTES):
+            chunk = payload[off:off + OBJ_CHUNK_BYTES]
+            dist.recv(chunk, src=src, group=self._group, tag=tag)
         return pickle.loads(payload.numpy().tobytes())
 
     # NOTE: these are hand-rolled (pickle once -> length-prefixed byte
@@ -146,12 +155,16 @@ class CommunicatorBase(object):
             payload = torch.frombuffer(bytearray(buf), dtype=torch.uint8)
             length = torch.tensor([payload.numel()], dtype=torch.long)
             dist.broadcast(length, src=src, group=self._group)
-            dist.broadcast(payload, src=src, group=self._group)
+            for off in range(0, max(payload.numel(), 1), OBJ_CHUNK_BYTES):
+                dist.broadcast(payload[off:off + OBJ_CHUNK_BYTES], src=src,
+                               group=self._group)
             return obj
         length = torch.zeros(1, dtype=torch.long)
         dist.broadcast(length, src=src, group=self._group)
         payload = torch.empty(int(length.item()), dtype=torch.uint8)
-        dist.broadcast(payload, src=src, group=self._group)
+        for off in range(0, max(payload.numel(), 1), OBJ_CHUNK_BYTES):
+            chunk = payload[off:off + OBJ_CHUNK_BYTES]
+            dist.broadcast(chunk, src=src, group=self._group)
         return pickle.loads(payload.numpy().tobytes())
 
     def gather_obj(self, obj, root=0):

@@ -210,3 +210,25 @@ def test_dummy_communicator():
 def test_unknown_name():
     with pytest.raises(ValueError):
         chainermn_amd.create_communicator('nonexistent')
+
+
+def _chunked_obj_worker(rank, size):
+    from chainermn_amd.communicators import _base
+    _base.OBJ_CHUNK_BYTES = 1000  # force multi-chunk transport
+    try:
+        comm = chainermn_amd.create_communicator('naive')
+        big = {'blob': bytes(range(256)) * 40, 'n': rank}  # ~10 KB
+        if rank == 0:
+            comm.send_obj(big, 1)
+        else:
+            got = comm.recv_obj(0)
+            assert got['blob'] == big['blob'] and got['n'] == 0
+        out = comm.bcast_obj(big if rank == 0 else None)
+        assert out['blob'] == big['blob']
+    finally:
+        _base.OBJ_CHUNK_BYTES = 1 << 30
+    return True
+
+
+def test_chunked_obj_transport():
+    run_distributed(_chunked_obj_worker, 2)
