@@ -45,10 +45,14 @@ def main():
         else torch.device('cpu')
     torch.manual_seed(7)
 
-    if args.mnbn:
+    if args.mnbn and use_gpu:
+        model = ResNet50(fused_ops=True, comm=comm)  # fused multi-node BN
+    elif args.mnbn:
         from chainermn_amd.links import MultiNodeBatchNormalization
         model = ResNet50(norm_layer=lambda c: MultiNodeBatchNormalization(
             c, comm))
+    elif use_gpu:
+        model = ResNet50(fused_ops=True)
     else:
         model = ResNet50()
     dtype = torch.bfloat16 if use_gpu else torch.float32
