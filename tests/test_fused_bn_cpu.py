@@ -65,7 +65,7 @@ def test_resnet_fused_ops_cpu_matches_plain():
     m1 = ResNet50(num_classes=10, fused_ops=True)
     torch.manual_seed(2)
     m2 = ResNet50(num_classes=10)
-    x = torch.randn(2, 3, 64, 64)
+    x = torch.randn(2, 3, 32, 32)
     y1 = m1(x)
     y2 = m2(x)
     assert torch.allclose(y1, y2, atol=1e-4), (y1 - y2).abs().max()
