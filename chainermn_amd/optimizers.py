@@ -179,6 +179,16 @@ class DoubleBufferingOptimizer(object):
         s['cur'] = prev
         return loss
 
+    # chainermn name: update()
+    def update(self, lossfun=None, *args, **kwargs):
+        if lossfun is not None:
+            loss = lossfun(*args, **kwargs)
+            self.zero_grad()
+            loss.backward()
+            self.step()
+            return loss
+        return self.step()
+
     def zero_grad(self, set_to_none=False):
         # Views must stay bound to the arena: never set_to_none.
         self._state['arena'].zero_()

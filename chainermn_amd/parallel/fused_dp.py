@@ -286,8 +286,13 @@ class FusedDataParallelSGD(object):
     def load_state_dict(self, sd):
         self._step = sd['step']
         self.state.master.copy_(sd['master'])
-        if self.v is not None and sd['v'] is not None:
+        if self.v is not None and sd.get('v') is not None:
             self.v.copy_(sd['v'])
+        # params mirror the (possibly loaded) master
+        if self.state.lowp is not None:
+            ops.cast_flat(self.state.master, self.state.lowp)
+        else:
+            pass  # params ARE master views
 
 
 class FusedDataParallelAdam(FusedDataParallelSGD):
@@ -316,6 +321,18 @@ class FusedDataParallelAdam(FusedDataParallelSGD):
                        step=self._adam_step, adamw=self.adamw,
                        grad_scale=1.0 / max(self.comm.size, 1),
                        p_lowp=self.state.lowp)
+
+    def state_dict(self):
+        sd = super().state_dict()
+        sd.update({'m': self.m, 'vv': self.vv,
+                   'adam_step': self._adam_step})
+        return sd
+
+    def load_state_dict(self, sd):
+        super().load_state_dict(sd)
+        self.m.copy_(sd['m'])
+        self.vv.copy_(sd['vv'])
+        self._adam_step = sd['adam_step']
 
 
 class FusedDataParallelLARS(FusedDataParallelSGD):

@@ -101,3 +101,24 @@ def test_bucketed_overlap_matches_serial():
         opt.step()
     for p, r in zip(results[0], ref.parameters()):
         assert torch.allclose(p, r.detach(), atol=1e-5)
+
+
+def test_fused_dp_state_roundtrip():
+    comm = chainermn_amd.create_communicator('dummy')
+    model = _model(3)
+    opt = FusedDataParallelSGD(model, comm, lr=0.1, momentum=0.9,
+                               overlap=False, grad_dtype=torch.float32)
+    for _ in range(2):
+        opt.zero_grad()
+        model(torch.randn(4, 6)).pow(2).mean().backward()
+        opt.step()
+    sd = {k: (v.clone() if torch.is_tensor(v) else v)
+          for k, v in opt.state_dict().items()}
+    w_after = [p.detach().clone() for p in model.parameters()]
+    # perturb then restore
+    opt.state.master.add_(1.0)
+    opt.load_state_dict(sd)
+    assert torch.allclose(opt.state.master,
+                          sd['master'])
+    for p, w in zip(model.parameters(), w_after):
+        assert torch.allclose(p.data, w, atol=1e-6)
