@@ -232,3 +232,21 @@ def _chunked_obj_worker(rank, size):
 
 def test_chunked_obj_transport():
     run_distributed(_chunked_obj_worker, 2)
+
+
+def _pure_rccl_cpu_worker(rank, size):
+    """pure_rccl's pack->allreduce->unpack logic on the CPU fallbacks
+    (fp32 wire; the GPU path is covered by the gpu-marked suite)."""
+    comm = chainermn_amd.create_communicator('pure_rccl',
+                                             allreduce_grad_dtype=None)
+    model = ExampleModel()
+    _set_grads(model, float(rank + 1))
+    comm.multi_node_mean_grad(model)
+    expect = sum(range(1, size + 1)) / size
+    for p in model.parameters():
+        assert torch.allclose(p.grad, torch.full_like(p.grad, expect))
+    return True
+
+
+def test_pure_rccl_cpu_logic():
+    run_distributed(_pure_rccl_cpu_worker, 2)

@@ -90,3 +90,20 @@ def test_synchronized_iterator_syncs_rng():
     results = run_distributed(_sync_iterator_worker, 2)
     assert results[0][0] == results[1][0] == [0, 1, 2]
     assert results[0][1] == results[1][1]
+
+
+def _mn_iterator_epochs_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    base = [torch.tensor([i]) for i in range(3)] if rank == 0 else None
+    it = create_multi_node_iterator(base, comm, rank_master=0)
+    seen = []
+    for epoch in range(2):
+        for b in it:
+            seen.append((epoch, int(b[0])))
+    return seen
+
+
+def test_multi_node_iterator_multiple_epochs():
+    results = run_distributed(_mn_iterator_epochs_worker, 2)
+    assert results[0] == results[1]
+    assert results[0] == [(0, 0), (0, 1), (0, 2), (1, 0), (1, 1), (1, 2)]

@@ -141,3 +141,29 @@ def test_multi_node_snapshot_dedup():
     # only the replica-set leader (rank 0) wrote its file
     files = [f for f in results[0] if f.startswith('snap_')]
     assert files == ['snap_2.r0'], results[0]
+
+
+def _evaluator_metrics_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    model = nn.Linear(3, 2)
+    with torch.no_grad():
+        model.weight.zero_()
+        model.bias.fill_(float(rank))
+    data = [(torch.zeros(4, 3),)]
+
+    def loss_fn(model, batch):
+        return model(batch[0]).sum()
+
+    def acc_metric(model, batch):
+        return float(rank + 1)  # rank0: 1, rank1: 2 -> mean 1.5
+
+    ev = chainermn_amd.create_multi_node_evaluator(
+        Evaluator(data, model, loss_fn=loss_fn,
+                  metrics={'accuracy': acc_metric}), comm)
+    r = ev.evaluate()
+    assert abs(r['validation/main/accuracy'] - 1.5) < 1e-9
+    return True
+
+
+def test_evaluator_metrics_averaged():
+    run_distributed(_evaluator_metrics_worker, 2)
