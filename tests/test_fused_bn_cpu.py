@@ -68,4 +68,4 @@ def test_resnet_fused_ops_cpu_matches_plain():
     x = torch.randn(2, 3, 32, 32)
     y1 = m1(x)
     y2 = m2(x)
-    assert torch.allclose(y1, y2, atol=1e-4), (y1 - y2).abs().max()
+    assert torch.allclose(y1, y2, atol=1e-3), (y1 - y2).abs().max()
