@@ -168,3 +168,42 @@ def test_multi_node_n_step_rnn():
                           rnn_a.weight_ih_l0.grad, atol=1e-5)
     assert torch.allclose(torch.from_numpy(results[1]['g']),
                           rnn_b.weight_ih_l0.grad, atol=1e-5)
+
+
+def _multicast_chainlist_worker(rank, size):
+    """rank 0 multicasts its output to ranks 1 and 2 (rank_out list);
+    backward sums both consumers' gradients at the producer."""
+    comm = chainermn_amd.create_communicator('naive')
+    torch.manual_seed(8)
+    f = nn.Linear(4, 4)
+    g1 = nn.Linear(4, 2)
+    g2 = nn.Linear(4, 2)
+    model = MultiNodeChainList(comm)
+    if rank == 0:
+        model.add_link(f, rank_in=None, rank_out=[1, 2])
+        torch.manual_seed(12)
+        x = torch.randn(3, 4)
+        out = model(x)
+        out.backward()
+        return {'gw': f.weight.grad}
+    else:
+        link = g1 if rank == 1 else g2
+        model.add_link(link, rank_in=0, rank_out=None)
+        y = model()
+        (float(rank) * y.pow(2).sum()).backward()
+        return {}
+
+
+def test_chainlist_multicast_rank_out():
+    results = run_distributed(_multicast_chainlist_worker, 3)
+    torch.manual_seed(8)
+    f = nn.Linear(4, 4)
+    g1 = nn.Linear(4, 2)
+    g2 = nn.Linear(4, 2)
+    torch.manual_seed(12)
+    x = torch.randn(3, 4)
+    h = f(x)
+    loss = 1.0 * g1(h).pow(2).sum() + 2.0 * g2(h).pow(2).sum()
+    loss.backward()
+    assert torch.allclose(torch.from_numpy(results[0]['gw']),
+                          f.weight.grad, atol=1e-5)

@@ -621,3 +621,26 @@ def test_mfma_gemm_bt_vs_matmul():
         err = (d.float() - ref.float()).abs().max()
         ref_mag = ref.float().abs().max()
         assert err <= 0.02 * ref_mag + 0.1, (M, N, K, err, ref_mag)
+
+
+@pytest.mark.parametrize('C', [12, 24, 40])
+def test_bn_odd_channel_counts(C):
+    """Channels-last reduce paths for C%4==0 but not power-of-two lane
+    mappings (non-pow2 cross-row trees, CPT=4/8 splits)."""
+    from chainermn_amd.links import FusedBatchNorm2d
+    torch.manual_seed(71)
+    x32 = torch.randn(6, C, 9, 9, device=DEV)
+    x = x32.to(torch.bfloat16).to(memory_format=torch.channels_last)
+    x.requires_grad_(True)
+    xr = x32.clone().requires_grad_(True)
+    bn = FusedBatchNorm2d(C).to(DEV)
+    ref = torch.nn.BatchNorm2d(C).to(DEV)
+    y = bn(x)
+    yr = ref(xr)
+    gy = torch.randn_like(yr)
+    y.backward(gy.to(torch.bfloat16).to(memory_format=torch.channels_last))
+    yr.backward(gy)
+    assert torch.allclose(y.float(), yr.detach(), atol=5e-2), C
+    assert torch.allclose(x.grad.float(), xr.grad, atol=5e-2), C
+    assert torch.allclose(bn.weight.grad.float(), ref.weight.grad,
+                          atol=0.5, rtol=5e-2), C
