@@ -132,8 +132,11 @@ class _BucketEngine(object):
             self.stream = torch.cuda.Stream()
             self.events = [torch.cuda.Event() for _ in self.buckets]
         self._reset()
-        for p in state.params:
-            p.register_post_accumulate_grad_hook(self._hook)
+        # NOTE: hooks live as long as the params; building a second
+        # optimizer over the same model is unsupported (as is re-arenaing
+        # params via a second _FlatState).
+        self._hook_handles = [p.register_post_accumulate_grad_hook(
+            self._hook) for p in state.params]
 
     def _reset(self):
         self.pending = [len(b[2]) for b in self.buckets]
@@ -141,6 +144,12 @@ class _BucketEngine(object):
 
     def _hook(self, p):
         bi = self.param_bucket[id(p)]
+        if bi < self.launched or self.pending[bi] <= 0:
+            raise RuntimeError(
+                'bucketed overlap saw a second gradient for a bucket that '
+                'already reduced: multi-backward gradient accumulation is '
+                'not supported with overlap="bucketed" — use '
+                'overlap=False or overlap=True (double buffering)')
         self.pending[bi] -= 1
         self._launch_ready_prefix()
 
