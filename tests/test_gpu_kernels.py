@@ -644,3 +644,23 @@ def test_bn_odd_channel_counts(C):
     assert torch.allclose(x.grad.float(), xr.grad, atol=5e-2), C
     assert torch.allclose(bn.weight.grad.float(), ref.weight.grad,
                           atol=0.5, rtol=5e-2), C
+
+
+def test_pure_rccl_mean_grad_async_world1():
+    """Side-stream async mean-grad API (pack+all-reduce on a caller
+    stream, unpack after event wait)."""
+    import chainermn_amd
+    import torch.nn as nn
+    comm = chainermn_amd.create_communicator(
+        'pure_rccl', allreduce_grad_dtype='bf16')
+    model = nn.Linear(40, 9).to(DEV)
+    for p in model.parameters():
+        p.grad = torch.randn_like(p)
+    before = [p.grad.clone() for p in model.parameters()]
+    stream = torch.cuda.Stream()
+    handle = comm.mean_grad_async(model, stream)
+    handle.unpack()
+    torch.cuda.synchronize()
+    for p, b in zip(model.parameters(), before):
+        assert torch.allclose(p.grad, b.to(torch.bfloat16).float(),
+                              atol=1e-2)
