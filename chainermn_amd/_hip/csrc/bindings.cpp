@@ -53,6 +53,7 @@ torch::Tensor bn_fwd(torch::Tensor x, torch::Tensor mean, torch::Tensor rstd,
                      torch::Tensor gamma, torch::Tensor beta);
 std::tuple<torch::Tensor, torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x);
 torch::Tensor gemm_bt_bf16(torch::Tensor a, torch::Tensor b);
+torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -98,6 +99,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "channels-last 3x3/s2/p1 maxpool forward (+window codes)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd,
         "gather-based maxpool backward (no atomics)");
+  m.def("conv3x3s1_fwd_bf16", &conv3x3s1_fwd_bf16,
+        "EXPERIMENTAL implicit-GEMM 3x3/s1/p1 conv forward (MFMA)");
   m.def("gemm_bt_bf16", &gemm_bt_bf16,
         "hand-written MFMA bf16 GEMM: D = A @ B^T (128x128x64 tile, "
         "glds staging, XOR-swizzled LDS, XCD-aware grid)");

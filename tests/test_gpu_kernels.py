@@ -664,3 +664,19 @@ def test_pure_rccl_mean_grad_async_world1():
     for p, b in zip(model.parameters(), before):
         assert torch.allclose(p.grad, b.to(torch.bfloat16).float(),
                               atol=1e-2)
+
+
+def test_conv3x3_implicit_gemm_fwd():
+    """EXPERIMENTAL implicit-GEMM 3x3 conv forward vs F.conv2d."""
+    torch.manual_seed(81)
+    for N, C, H, W, Cout in [(2, 64, 16, 16, 128), (1, 128, 32, 32, 128),
+                             (2, 64, 32, 16, 256)]:
+        x = torch.randn(N, C, H, W, device=DEV, dtype=torch.bfloat16
+                        ).to(memory_format=torch.channels_last)
+        w = torch.randn(Cout, C, 3, 3, device=DEV, dtype=torch.bfloat16
+                        ).to(memory_format=torch.channels_last) * 0.05
+        y = _ext().conv3x3s1_fwd_bf16(x, w)
+        ref = torch.nn.functional.conv2d(x, w, padding=1)
+        err = (y.float() - ref.float()).abs().max()
+        mag = ref.float().abs().max()
+        assert err <= 0.03 * mag + 0.05, (N, C, H, W, Cout, err, mag)
