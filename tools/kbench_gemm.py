@@ -18,3 +18,15 @@ for M, N, K in [(1024,1024,1024),(4096,4096,4096),(8192,8192,8192),
     dt1 = t(lambda: ops.require().gemm_bt_bf16(a, b))
     dt2 = t(lambda: a @ b.t())
     print(f"  M{M} N{N} K{K}: ours {fl/dt1/1e12:7.1f} TF | hipBLASLt {fl/dt2/1e12:7.1f} TF")
+
+print("\nEXPERIMENTAL implicit-GEMM 3x3/s1/p1 conv fwd vs torch/MIOpen:")
+import torch.nn.functional as F
+for N, C, H, W, Cout in [(512, 128, 28, 28, 128), (512, 256, 14, 14, 256),
+                         (512, 512, 7, 7, 512)]:
+    if (N*H*W) % 128: continue
+    x = (torch.rand(N, C, H, W, device='cuda', dtype=torch.bfloat16)*2-1).to(memory_format=torch.channels_last)
+    w = ((torch.rand(Cout, C, 3, 3, device='cuda', dtype=torch.bfloat16)*2-1)*0.05).to(memory_format=torch.channels_last)
+    fl = 2.0*N*H*W*Cout*9*C
+    d1 = t(lambda: ops.require().conv3x3s1_fwd_bf16(x, w))
+    d2 = t(lambda: F.conv2d(x, w, padding=1))
+    print(f"  N{N} C{C} {H}x{W} -> {Cout}: ours {fl/d1/1e12:6.1f} TF ({d1*1e6:6.1f}us) | miopen {fl/d2/1e12:6.1f} TF ({d2*1e6:6.1f}us)")
