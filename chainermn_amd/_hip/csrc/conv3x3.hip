@@ -42,14 +42,9 @@ __global__ __launch_bounds__(C3_THREADS) void conv3x3s1_fwd_kernel(
     const short* __restrict__ Wt, // [Cout, 3, 3, Cin]
     short* __restrict__ Y,        // [N, H, W, Cout]
     int Nb, int H, int Wd, int Cin, int Cout) {
-  __shared__ __attribute__((aligned(16))) char lds[2 * 2 * C3_BM * C3_BK
-                                                   * 2];
-  auto ldsA = [&](int buf) -> char* {
-    return lds + buf * 2 * C3_BM * C3_BK * 2;
-  };
-  auto ldsB = [&](int buf) -> char* {
-    return lds + buf * 2 * C3_BM * C3_BK * 2 + C3_BM * C3_BK * 2;
-  };
+  __shared__ __attribute__((aligned(16))) char lds[2 * C3_BM * C3_BK * 2];
+  char* ldsA = lds;
+  char* ldsB = lds + C3_BM * C3_BK * 2;
 
   const long M = (long)Nb * H * Wd;
   const int ntile_n = Cout / C3_BN;
@@ -68,13 +63,14 @@ __global__ __launch_bounds__(C3_THREADS) void conv3x3s1_fwd_kernel(
 #pragma unroll
     for (int n = 0; n < 4; ++n) acc[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // staging helper: A via predicated register loads + swizzled ds_write
-  // (im2col with padding), B via glds with pre-swizzled source
-  auto stage = [&](int buf, int kt) {
+  const int nkt = KW / C3_BK;
+  for (int kt = 0; kt < nkt; ++kt) {
     const int k0 = kt * C3_BK;
     const int r = k0 / Cin;          // tap index 0..8
     const int ky = r / 3, kx = r % 3;
     const int ci0 = k0 - r * Cin;
+    // ---- stage A: register loads with boundary predication, swizzled
+    // ds_write (2 chunks of 16 B per thread cover 128x64 bf16) ----
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int p = (i * C3_THREADS + t) * 16;
@@ -95,9 +91,9 @@ __global__ __launch_bounds__(C3_THREADS) void conv3x3s1_fwd_kernel(
           v = *reinterpret_cast<const uint4*>(src);
         }
       }
-      *reinterpret_cast<uint4*>(ldsA(buf) + row * 128
-                                + c3_swz(row, within)) = v;
+      *reinterpret_cast<uint4*>(ldsA + row * 128 + c3_swz(row, within)) = v;
     }
+    // ---- stage B: weights [Cout][9][Cin]; row = co within tile ----
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       // glds needs a WAVE-UNIFORM LDS base (+ lane*16 implicit)
@@ -110,17 +106,10 @@ __global__ __launch_bounds__(C3_THREADS) void conv3x3s1_fwd_kernel(
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)
-              (ldsB(buf) + base),
+              (ldsB + base),
           16, 0, 0);
     }
-  };
-
-  const int nkt = KW / C3_BK;
-  stage(0, 0);
-  __syncthreads();
-  for (int kt = 0; kt < nkt; ++kt) {
-    const int buf = kt & 1;
-    if (kt + 1 < nkt) stage(buf ^ 1, kt + 1);
+    __syncthreads();
     const int fr = lane & 15;
     const int kq = lane >> 4;
 #pragma unroll
@@ -130,15 +119,13 @@ __global__ __launch_bounds__(C3_THREADS) void conv3x3s1_fwd_kernel(
       for (int m = 0; m < 4; ++m) {
         const int row = wr * 64 + m * 16 + fr;
         const int kbyte = ks * 64 + kq * 16;
-        a_frag[m] = *(const bf16x8*)(ldsA(buf) + row * 128
-                                     + c3_swz(row, kbyte));
+        a_frag[m] = *(const bf16x8*)(ldsA + row * 128 + c3_swz(row, kbyte));
       }
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
         const int row = wc * 64 + n * 16 + fr;
         const int kbyte = ks * 64 + kq * 16;
-        b_frag[n] = *(const bf16x8*)(ldsB(buf) + row * 128
-                                     + c3_swz(row, kbyte));
+        b_frag[n] = *(const bf16x8*)(ldsB + row * 128 + c3_swz(row, kbyte));
       }
 #pragma unroll
       for (int m = 0; m < 4; ++m)
