@@ -14,7 +14,12 @@
 // predication (padding rows/cols load zeros) and written to LDS at the
 // XOR-swizzled address directly (register staging may scatter, unlike
 // global_load_lds); the weight tile uses glds with a pre-swizzled source
-// like gemm.hip.  Correctness-first: single-buffered LDS, no pipelining.
+// like gemm.hip.  Single-buffered: measured 280-425 TF vs MIOpen's ~460;
+// a naive 2-buffer in-loop staging variant measured SLOWER (ordinary
+// register loads beside in-flight glds hit hipcc's vmcnt(0) drain — the
+// programming guide's mixed-load-kinds trap), so the pipelined version
+// needs the T14 split (issue loads early, ds_write after the barrier) —
+// round-2 work, see STATUS.md.
 //
 // Not wired into any model path; exposed as ops-level experimental API.
 
