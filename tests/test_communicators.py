@@ -250,3 +250,38 @@ def _pure_rccl_cpu_worker(rank, size):
 
 def test_pure_rccl_cpu_logic():
     run_distributed(_pure_rccl_cpu_worker, 2)
+
+
+def _split_of_split_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    half = comm.split(color=rank // 2, key=rank)    # {0,1} and {2,3}
+    assert half.size == 2
+    solo = half.split(color=half.rank, key=0)       # singletons
+    assert solo.size == 1 and solo.rank == 0
+    out = half.allreduce(torch.tensor([float(rank)]))
+    mates = [r for r in range(size) if r // 2 == rank // 2]
+    assert out.item() == float(sum(mates))
+    return True
+
+
+def test_split_of_split():
+    run_distributed(_split_of_split_worker, 4)
+
+
+def _pure_rccl_split_worker(rank, size):
+    comm = chainermn_amd.create_communicator('pure_rccl',
+                                             allreduce_grad_dtype=None)
+    sub = comm.split(color=rank % 2, key=rank)
+    assert type(sub).__name__ == 'PureRcclCommunicator'
+    model = ExampleModel()
+    _set_grads(model, float(rank + 1))
+    sub.multi_node_mean_grad(model)
+    mates = [r + 1 for r in range(size) if r % 2 == rank % 2]
+    expect = sum(mates) / len(mates)
+    for p in model.parameters():
+        assert torch.allclose(p.grad, torch.full_like(p.grad, expect))
+    return True
+
+
+def test_pure_rccl_split_preserves_class():
+    run_distributed(_pure_rccl_split_worker, 4)

@@ -177,3 +177,18 @@ def _pseudo_connect_worker(rank, size):
 
 def test_pseudo_connect():
     run_distributed(_pseudo_connect_worker, 2)
+
+
+def _allreduce_fn_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    x = torch.full((3,), float(rank + 1), requires_grad=True)
+    y = functions.allreduce(comm, x)
+    assert torch.allclose(y, torch.full((3,), 3.0))  # 1 + 2
+    ((rank + 1.0) * y.sum()).backward()
+    # dy/dx gets all-reduced too: sum of per-rank gy = 1 + 2 = 3
+    assert torch.allclose(x.grad, torch.full((3,), 3.0)), x.grad
+    return True
+
+
+def test_allreduce_function_grad():
+    run_distributed(_allreduce_fn_worker, 2)
