@@ -123,23 +123,26 @@ class CommunicatorBase(object):
         return dist.get_global_rank(self._group, group_rank)
 
     # -- object channel (reference: mpi_communicator_base *_obj, ~L80-150) -
-    def send_obj(self, obj, dest, tag=0):
+    def send_obj(self, obj, dest, tag=0, max_buf_len=None):
+        chunk_bytes = max_buf_len or OBJ_CHUNK_BYTES
         buf = pickle.dumps(obj, protocol=pickle.HIGHEST_PROTOCOL)
         payload = torch.frombuffer(bytearray(buf), dtype=torch.uint8)
-        length = torch.tensor([payload.numel()], dtype=torch.long)
+        length = torch.tensor([payload.numel(), chunk_bytes],
+                              dtype=torch.long)
         dst = self._global_rank(dest)
         dist.send(length, dst=dst, group=self._group, tag=tag)
-        for off in range(0, max(payload.numel(), 1), OBJ_CHUNK_BYTES):
-            dist.send(payload[off:off + OBJ_CHUNK_BYTES], dst=dst,
+        for off in range(0, max(payload.numel(), 1), chunk_bytes):
+            dist.send(payload[off:off + chunk_bytes], dst=dst,
                       group=self._group, tag=tag)
 
     def recv_obj(self, source, tag=0):
         src = self._global_rank(source)
-        length = torch.zeros(1, dtype=torch.long)
+        length = torch.zeros(2, dtype=torch.long)
         dist.recv(length, src=src, group=self._group, tag=tag)
-        payload = torch.empty(int(length.item()), dtype=torch.uint8)
-        for off in range(0, max(payload.numel(), 1), OBJ_CHUNK_BYTES):
-            chunk = payload[off:off + OBJ_CHUNK_BYTES]
+        chunk_bytes = int(length[1].item())
+        payload = torch.empty(int(length[0].item()), dtype=torch.uint8)
+        for off in range(0, max(payload.numel(), 1), chunk_bytes):
+            chunk = payload[off:off + chunk_bytes]
             dist.recv(chunk, src=src, group=self._group, tag=tag)
         return pickle.loads(payload.numpy().tobytes())
 
@@ -180,14 +183,15 @@ class CommunicatorBase(object):
         gathered = self.gather_obj(obj, root=0)
         return self.bcast_obj(gathered, root=0)
 
-    def scatter_obj(self, objs, root=0):
+    def scatter_obj(self, objs, root=0, max_buf_len=None):
         if self._rank == root:
             objs = list(objs)
             if len(objs) != self._size:
                 raise ValueError('scatter_obj requires len(objs) == size')
             for r in range(self._size):
                 if r != root:
-                    self.send_obj(objs[r], r, tag=8)
+                    self.send_obj(objs[r], r, tag=8,
+                                  max_buf_len=max_buf_len)
             return objs[root]
         return self.recv_obj(root, tag=8)
 

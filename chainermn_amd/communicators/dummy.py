@@ -31,7 +31,7 @@ class DummyCommunicator(CommunicatorBase):
     def bcast_data(self, model, root=0):
         pass
 
-    def send_obj(self, obj, dest, tag=0):
+    def send_obj(self, obj, dest, tag=0, max_buf_len=None):
         raise RuntimeError('DummyCommunicator has no peers')
 
     def recv_obj(self, source, tag=0):
@@ -46,7 +46,7 @@ class DummyCommunicator(CommunicatorBase):
     def allgather_obj(self, obj):
         return [obj]
 
-    def scatter_obj(self, objs, root=0):
+    def scatter_obj(self, objs, root=0, max_buf_len=None):
         return objs[0]
 
     def allreduce_obj(self, obj):

@@ -36,7 +36,7 @@ def scatter_dataset(dataset, comm, root=0, shuffle=False, seed=None,
             order = numpy.arange(n)
         slices = _split(order, comm.size, force_equal_length)
         shards = [[dataset[int(i)] for i in sl] for sl in slices]
-        return comm.scatter_obj(shards, root=root)
+        return comm.scatter_obj(shards, root=root, max_buf_len=max_buf_len)
     else:
         return comm.scatter_obj(None, root=root)
 
