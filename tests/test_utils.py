@@ -47,3 +47,12 @@ def test_pack_plan_alignment():
     assert plan.total == 128 + 192  # 129 rounded up to 192
     for off in plan.offsets_cpu:
         assert off % ops.ALIGN == 0
+
+
+def test_gemm_wrappers_cpu_fallback():
+    import torch
+    a = torch.randn(32, 16)
+    b = torch.randn(24, 16)
+    assert torch.allclose(ops.gemm_bt(a, b), a @ b.t(), atol=1e-5)
+    c = torch.randn(16, 24)
+    assert torch.allclose(ops.gemm_nn(a, c), a @ c, atol=1e-5)
