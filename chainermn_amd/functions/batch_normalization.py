@@ -18,9 +18,7 @@ from chainermn_amd import ops
 
 
 def _local_sums(x):
-    """(sum_c, sqsum_c, count) reduced over every dim but dim 1."""
-    if x.is_cuda:
-        return ops.require().bn_stats(x)
+    """(sum_c, sqsum_c, count) reduced over every dim but dim 1 (CPU)."""
     dims = [d for d in range(x.dim()) if d != 1]
     xf = x.float()
     return (xf.sum(dim=dims), (xf * xf).sum(dim=dims),
@@ -75,20 +73,15 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
         x, gamma, mean, rstd, total = ctx.saved_tensors
         comm = ctx.comm
         C = x.shape[1]
+        buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
         if x.is_cuda:
-            buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
             ops.require().bn_bwd_reduce_into(x, gy, mean, rstd, buf)
-            gbeta = buf[:C]
-            ggamma = buf[C:]
         else:
             dims = [d for d in range(x.dim()) if d != 1]
             gyf = gy.float()
             xhat = (x.float() - _bcast_shape(mean, x)) * _bcast_shape(rstd, x)
-            gbeta = gyf.sum(dim=dims)
-            ggamma = (gyf * xhat).sum(dim=dims)
-        buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
-        buf[:C] = gbeta
-        buf[C:] = ggamma
+            buf[:C] = gyf.sum(dim=dims)
+            buf[C:] = (gyf * xhat).sum(dim=dims)
         if comm.size > 1:
             buf = comm.allreduce(buf)
         gbeta_g = buf[:C]
