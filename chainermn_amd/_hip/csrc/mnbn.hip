@@ -137,11 +137,16 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
   const int tr = threadIdx.x / ct;
   for (int c_base = 0; c_base < C; c_base += tile_ch) {
     const int c = c_base + tc * CPT;
+    // Last tile when C > ct*CPT: threads whose channel slice starts past C
+    // must not touch memory (the caller guarantees C % CPT == 0, so
+    // c < C implies the whole [c, c+CPT) slice is in bounds). They still
+    // write their zeroed accumulator into LDS for the cross-row tree.
+    const bool in_c = c < C;
     float a[2 * CPT];
 #pragma unroll
     for (int k = 0; k < 2 * CPT; ++k) a[k] = 0.f;
     float m4[CPT], r4[CPT], g4[CPT], b4[CPT];
-    if (BWD) {
+    if (BWD && in_c) {
 #pragma unroll
       for (int k = 0; k < CPT; ++k) {
         m4[k] = mean[c + k];
@@ -153,7 +158,7 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
       }
     }
     const long row_stride = (long)gridDim.x * rpb;
-    if (tr < rpb) {
+    if (tr < rpb && in_c) {
       for (long m = blockIdx.x * (long)rpb + tr; m < M; m += row_stride) {
         const long idx = m * C + c;
         float xv[CPT], gv[CPT], yv[CPT];
@@ -196,7 +201,7 @@ __global__ void bn_reduce_cl_kernel(const T* __restrict__ x,
       __syncthreads();
       rem = off;
     }
-    if (tr == 0) {
+    if (tr == 0 && in_c) {
       // per-block partials: no same-address atomic chains
 #pragma unroll
       for (int k = 0; k < CPT; ++k) {

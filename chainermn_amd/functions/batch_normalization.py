@@ -66,6 +66,11 @@ class _MultiNodeBatchNorm(torch.autograd.Function):
                  + _bcast_shape(beta.float(), x)).to(x.dtype)
         ctx.save_for_backward(x, gamma, mean, rstd, total)
         ctx.comm = comm
+        # The returned statistics are a by-product: backward ignores
+        # g_mean/g_var, so a user graph built through them would silently
+        # get zero gradient.  Marking them non-differentiable makes such
+        # misuse raise instead.
+        ctx.mark_non_differentiable(mean, var)
         return y, mean, var
 
     @staticmethod

@@ -9,7 +9,8 @@ LARS).
 Implementation: the fused CDNA4 BN engine (fused_batch_norm._FusedBNAct)
 with a communicator attached — the (Σx, Σx², Σgy', Σgy'x̂) buffers go over
 RCCL between the local reduction and the finalize/input kernels.  Running
-statistics use the reference's convention (decay, biased batch variance).
+statistics use the reference's convention (decay; unbiased m/(m-1)
+running-variance adjust, with m = the global element count).
 ``MultiNodeBNReLU2d`` / ``forward_add_relu`` give the multi-node BN the
 same ReLU / residual fusions as the local FusedBatchNorm2d.
 """
@@ -22,7 +23,11 @@ from chainermn_amd.links.fused_batch_norm import FusedBatchNorm2d
 
 class MultiNodeBatchNormalization(FusedBatchNorm2d):
 
-    _running_unbias = 1.0  # reference keeps biased batch var
+    # Upstream Chainer/ChainerMN BN applies the unbiased m/(m-1) adjust
+    # when folding batch var into running_var; None selects the engine's
+    # n/(n-1) with n = the GLOBAL element count (the statistics the var
+    # was computed over), matching that convention.
+    _running_unbias = None
 
     def __init__(self, size, comm, decay=0.9, eps=2e-5, dtype=None,
                  use_gamma=True, use_beta=True, communication_backend='auto'):

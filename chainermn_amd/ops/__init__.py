@@ -226,6 +226,34 @@ def _time_gpu(fn, n=5):
     return (time.perf_counter() - t0) / n
 
 
+def _decide(key, mfma_fn, blas_fn):
+    """Pick mfma-vs-blas for one shape, identically on every rank.
+
+    Rank 0 times both candidates (this calls torch.cuda.synchronize, so
+    the one-time cost per new shape stalls any overlapped side-stream
+    work once); the other ranks receive the choice over the CPU object
+    channel instead of timing locally — local timing jitter would let
+    ranks pick different kernels and make activations rank-divergent.
+    All ranks reach this point at the same place in the same order (same
+    model, same step), so the broadcast is aligned.
+    """
+    import torch.distributed as dist
+    synced = (dist.is_available() and dist.is_initialized()
+              and dist.get_world_size() > 1)
+    if synced and dist.get_rank() != 0:
+        lst = [None]
+        dist.broadcast_object_list(lst, src=0)
+        choice = lst[0]
+    else:
+        t_ours = _time_gpu(mfma_fn)
+        t_blas = _time_gpu(blas_fn)
+        choice = 'mfma' if t_ours < t_blas else 'blas'
+        if synced:
+            dist.broadcast_object_list([choice], src=0)
+    _GEMM_CHOICE[key] = choice
+    return choice
+
+
 def gemm_bt(a, b):
     """D = a @ b.T.  Eligible bf16 shapes (M%128 == N%128 == 0, K%64 == 0)
     dispatch to whichever of {hand-written MFMA kernel, hipBLASLt} measured
@@ -242,10 +270,8 @@ def gemm_bt(a, b):
     choice = _GEMM_CHOICE.get(key)
     if choice is None:
         ext = require()
-        t_ours = _time_gpu(lambda: ext.gemm_bt_bf16(a, b))
-        t_blas = _time_gpu(lambda: a @ b.t())
-        choice = 'mfma' if t_ours < t_blas else 'blas'
-        _GEMM_CHOICE[key] = choice
+        choice = _decide(key, lambda: ext.gemm_bt_bf16(a, b),
+                         lambda: a @ b.t())
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b)
     return a @ b.t()
@@ -264,11 +290,9 @@ def gemm_nn(a, b):
     choice = _GEMM_CHOICE.get(key)
     if choice is None:
         ext = require()
-        t_ours = _time_gpu(
-            lambda: ext.gemm_bt_bf16(a, b.t().contiguous()))
-        t_blas = _time_gpu(lambda: a @ b)
-        choice = 'mfma' if t_ours < t_blas else 'blas'
-        _GEMM_CHOICE[key] = choice
+        choice = _decide(key,
+                         lambda: ext.gemm_bt_bf16(a, b.t().contiguous()),
+                         lambda: a @ b)
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b.t().contiguous())
     return a @ b

@@ -27,6 +27,32 @@ import torch.distributed as dist
 from chainermn_amd import ops
 
 
+def _is_cl(t):
+    return (t.dim() == 4 and not t.is_contiguous()
+            and t.is_contiguous(memory_format=torch.channels_last))
+
+
+def _arena_view(flat, like):
+    """View a flat arena slice with ``like``'s shape AND memory format.
+
+    channels_last params keep their strides (conv kernel selection on
+    MI355X depends on it); the arena then holds the param's PHYSICAL
+    (NHWC) element order, which is also what ``_physical_flat`` yields.
+    """
+    if _is_cl(like):
+        n, c, h, w = like.shape
+        return flat.as_strided(like.shape, (h * w * c, 1, w * c, c))
+    return flat.view_as(like)
+
+
+def _physical_flat(t):
+    """``t``'s elements in memory order — the arena layout of
+    ``_arena_view`` — regardless of memory format."""
+    if _is_cl(t):
+        return t.permute(0, 2, 3, 1).reshape(-1)
+    return t.reshape(-1)
+
+
 class _FlatState(object):
     """Flattens a model's params into arenas and rebinds p.data/p.grad."""
 
@@ -67,9 +93,10 @@ class _FlatState(object):
         with torch.no_grad():
             for p, off, sz in zip(params, self.plan.offsets_cpu,
                                   self.plan.sizes_cpu):
-                self.master[off:off + sz].copy_(p.data.reshape(-1).float())
+                self.master[off:off + sz].copy_(
+                    _physical_flat(p.data).float())
                 arena = self.lowp if self.lowp is not None else self.master
-                view = arena[off:off + sz].view_as(p.data)
+                view = _arena_view(arena[off:off + sz], p.data)
                 view.copy_(p.data)
                 p.data = view
         self.cur = 0
@@ -78,7 +105,7 @@ class _FlatState(object):
     def _bind_grads(self, idx):
         for p, off, sz in zip(self.params, self.plan.offsets_cpu,
                               self.plan.sizes_cpu):
-            p.grad = self.grads[idx][off:off + sz].view_as(p.data)
+            p.grad = _arena_view(self.grads[idx][off:off + sz], p.data)
 
     def grad_arena(self, idx=None):
         return self.grads[self.cur if idx is None else idx]
@@ -89,7 +116,8 @@ class _FlatState(object):
         with torch.no_grad():
             for p, off, sz in zip(self.params, self.plan.offsets_cpu,
                                   self.plan.sizes_cpu):
-                self.master[off:off + sz].copy_(p.data.reshape(-1).float())
+                self.master[off:off + sz].copy_(
+                    _physical_flat(p.data).float())
 
 
 class _BucketEngine(object):
@@ -142,6 +170,15 @@ class _BucketEngine(object):
         self.pending = [len(b[2]) for b in self.buckets]
         self.launched = 0  # buckets [0, launched) already issued
 
+    # --- overlap telemetry -------------------------------------------
+    # If backward completion order diverges from reverse arena order
+    # (branchy models), ready buckets pile up behind a not-yet-ready
+    # prefix bucket and flush synchronously at finish() — correct but
+    # with no overlap.  Count it and warn once so the degradation is
+    # visible instead of silent.
+    flushed_at_finish = 0
+    _warned_flush = False
+
     def _hook(self, p):
         bi = self.param_bucket[id(p)]
         if bi < self.launched or self.pending[bi] <= 0:
@@ -174,6 +211,20 @@ class _BucketEngine(object):
 
     def finish(self):
         """Flush un-launched buckets, wait for all; call before update."""
+        n_flush = len(self.buckets) - self.launched
+        if n_flush > 0:
+            self.flushed_at_finish += n_flush
+            if not self._warned_flush and len(self.buckets) > 1:
+                self._warned_flush = True
+                import warnings
+                warnings.warn(
+                    'bucketed overlap: %d of %d buckets flushed '
+                    'synchronously at finish() (backward completion '
+                    'order diverges from arena order, or some params '
+                    'got no gradient) — training is correct but those '
+                    'buckets did not overlap with backward; see '
+                    '_BucketEngine.flushed_at_finish' %
+                    (n_flush, len(self.buckets)))
         for bi in range(self.launched, len(self.buckets)):
             self.pending[bi] = 0
         self._launch_ready_prefix()
@@ -223,10 +274,23 @@ class FusedDataParallelSGD(object):
         self._step = 0
 
     def zero_grad(self, set_to_none=False):
-        # current arena is zeroed inside step(); only needed before the
-        # very first backward or after manual grad fiddling
+        """No-op after step 0 BY DESIGN: step() zeroes the active arena
+        after applying it, and p.grad must stay bound to the arena views
+        (``set_to_none=True`` would break the zero-copy wire path, so it
+        raises).  If you mutate ``p.grad`` manually between steps, call
+        ``zero_grad_force()`` — plain zero_grad will NOT clear the
+        arena for you."""
+        if set_to_none:
+            raise ValueError(
+                'FusedDataParallel binds p.grad to the flat wire arena; '
+                'set_to_none=True is unsupported (pass False)')
         if self._step == 0:
             self.state.grad_arena().zero_()
+
+    def zero_grad_force(self):
+        """Explicitly zero the ACTIVE grad arena (after manual p.grad
+        mutation; normal training never needs this)."""
+        self.state.grad_arena().zero_()
 
     def _reduce_async(self, idx):
         g = self.state.grads[idx]

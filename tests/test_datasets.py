@@ -107,3 +107,55 @@ def test_multi_node_iterator_multiple_epochs():
     results = run_distributed(_mn_iterator_epochs_worker, 2)
     assert results[0] == results[1]
     assert results[0] == [(0, 0), (0, 1), (0, 2), (1, 0), (1, 1), (1, 2)]
+
+
+def _mn_iterator_epoch_state_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    base = [torch.tensor([i]) for i in range(3)] if rank == 0 else None
+    it = create_multi_node_iterator(base, comm, rank_master=0)
+    trace = []
+    for _ in range(2):
+        for b in it:
+            trace.append((it.epoch, it.is_new_epoch, int(b[0])))
+        trace.append(('end', it.epoch))
+    return trace
+
+
+def test_multi_node_iterator_epoch_state_on_slaves():
+    """Slaves see per-batch epoch/is_new_epoch identical to the master
+    (epoch-triggered extensions fire on the same iteration everywhere)."""
+    results = run_distributed(_mn_iterator_epoch_state_worker, 2)
+    assert results[0] == results[1]
+    assert results[0] == [
+        (0, False, 0), (0, False, 1), (0, False, 2), ('end', 1),
+        (1, True, 0), (1, False, 1), (1, False, 2), ('end', 2)]
+
+
+def _sync_iterator_reseed_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    torch.manual_seed(100 + rank)  # deliberately different
+
+    class Shuffler:
+        def __iter__(self):
+            return iter(torch.randperm(16).tolist())
+
+        def __len__(self):
+            return 16
+
+    it = create_synchronized_iterator(Shuffler(), comm)
+    epochs = []
+    for _ in range(3):
+        epochs.append(list(it))
+        # consume a DIFFERENT amount of randomness per rank between
+        # epochs — a one-time seed sync would diverge here
+        for _ in range(rank + 1):
+            torch.rand(5)
+    return epochs
+
+
+def test_synchronized_iterator_resyncs_every_epoch():
+    results = run_distributed(_sync_iterator_reseed_worker, 2)
+    assert results[0] == results[1]
+    assert len(results[0]) == 3
+    # shuffles differ across epochs (the RNG advances on rank 0)
+    assert results[0][0] != results[0][1]

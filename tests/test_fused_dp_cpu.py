@@ -122,3 +122,50 @@ def test_fused_dp_state_roundtrip():
                           sd['master'])
     for p, w in zip(model.parameters(), w_after):
         assert torch.allclose(p.data, w, atol=1e-6)
+
+
+def test_fused_dp_preserves_channels_last():
+    """Wrapping a channels_last model must keep conv-weight memory format
+    (kernel selection on MI355X depends on it) AND train identically to
+    the same model in contiguous format (ADVICE.md round-1 finding)."""
+    comm = chainermn_amd.create_communicator('dummy')
+
+    def conv_model(seed):
+        torch.manual_seed(seed)
+        return nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+                             nn.Conv2d(8, 4, 1))
+
+    model = conv_model(11).to(memory_format=torch.channels_last)
+    opt = FusedDataParallelSGD(model, comm, lr=0.05, momentum=0.9,
+                               overlap=False, grad_dtype=torch.float32)
+    for p in model.parameters():
+        if p.dim() == 4:
+            assert p.data.is_contiguous(
+                memory_format=torch.channels_last), p.shape
+            if p.shape[2] * p.shape[3] > 1:  # 1x1: CL ≡ contiguous
+                assert not p.data.is_contiguous(), p.shape
+            assert p.grad.is_contiguous(
+                memory_format=torch.channels_last), p.shape
+
+    ref = conv_model(11)
+    ropt = torch.optim.SGD(ref.parameters(), lr=0.05, momentum=0.9)
+    for step in range(3):
+        torch.manual_seed(90 + step)
+        x = torch.randn(2, 3, 8, 8)
+        opt.zero_grad()
+        model(x.to(memory_format=torch.channels_last)
+              ).pow(2).mean().backward()
+        opt.step()
+        ropt.zero_grad()
+        ref(x).pow(2).mean().backward()
+        ropt.step()
+    for p, r in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, r.detach(), atol=1e-5), \
+            (p.data - r.detach()).abs().max()
+    # state roundtrip keeps the format too
+    sd = {k: (v.clone() if torch.is_tensor(v) else v)
+          for k, v in opt.state_dict().items()}
+    opt.load_state_dict(sd)
+    for p in model.parameters():
+        if p.dim() == 4:
+            assert p.data.is_contiguous(memory_format=torch.channels_last)

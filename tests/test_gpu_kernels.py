@@ -300,6 +300,7 @@ def test_fused_dp_overlap_staleness_semantics():
     model = nn.Linear(8, 8).to(DEV)
     ref = nn.Linear(8, 8).to(DEV)
     ref.load_state_dict(model.state_dict())
+    init_sd = {k: v.clone() for k, v in model.state_dict().items()}
     opt = FusedDataParallelSGD(model, comm, lr=0.1, momentum=0.0,
                                overlap=True)
     ropt = torch.optim.SGD(ref.parameters(), lr=0.1)
@@ -312,20 +313,38 @@ def test_fused_dp_overlap_staleness_semantics():
     opt.zero_grad()
     opt.step()
     torch.cuda.synchronize()
+    # exact 1-step-stale reference: step k computes g_k at the CURRENT
+    # params, THEN applies g_{k-1} (fp32 arenas, momentum 0 → bitwise-
+    # comparable update order)
+    pending = None
     for x in xs:
         ropt.zero_grad()
         ref(x).pow(2).mean().backward()
-        ropt.step()
-    # after flush, overlapped run has applied all 4 gradients, but each was
-    # computed on the params one step older — only the FIRST update used
-    # identical params, so compare against a manual stale recomputation
-    stale = nn.Linear(8, 8).to(DEV)
-    stale.load_state_dict({k: v for k, v in zip(
-        ref.state_dict().keys(),
-        [t.clone() for t in model.state_dict().values()])})
-    # semantic check: params changed and loss decreases
-    l0 = model(xs[0]).pow(2).mean()
-    assert torch.isfinite(l0)
+        g = [p.grad.clone() for p in ref.parameters()]
+        if pending is not None:
+            with torch.no_grad():
+                for p, gp in zip(ref.parameters(), pending):
+                    p.sub_(gp, alpha=0.1)
+        pending = g
+    with torch.no_grad():
+        for p, gp in zip(ref.parameters(), pending):
+            p.sub_(gp, alpha=0.1)
+    for p, r in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, r.data, atol=1e-6), \
+            (p.data - r.data).abs().max()
+    # and the stale trajectory must differ from the synchronous one —
+    # otherwise this test would pass vacuously on a non-overlapped
+    # implementation
+    sync = nn.Linear(8, 8).to(DEV)
+    sync.load_state_dict(init_sd)
+    sopt = torch.optim.SGD(sync.parameters(), lr=0.1)
+    for x in xs:
+        sopt.zero_grad()
+        sync(x).pow(2).mean().backward()
+        sopt.step()
+    diff = max((p.data - s.data).abs().max().item()
+               for p, s in zip(model.parameters(), sync.parameters()))
+    assert diff > 1e-5, diff
 
 
 def test_bench_smoke():
@@ -623,10 +642,15 @@ def test_mfma_gemm_bt_vs_matmul():
         assert err <= 0.02 * ref_mag + 0.1, (M, N, K, err, ref_mag)
 
 
-@pytest.mark.parametrize('C', [12, 24, 40])
+@pytest.mark.parametrize('C', [12, 24, 40, 1028, 2560])
 def test_bn_odd_channel_counts(C):
     """Channels-last reduce paths for C%4==0 but not power-of-two lane
-    mappings (non-pow2 cross-row trees, CPT=4/8 splits)."""
+    mappings (non-pow2 cross-row trees, CPT=4/8 splits).
+
+    C=1028 (CPT=4, C/CPT=257>256) and C=2560 (CPT=8, C/CPT=320>256)
+    force the channel-tiled path of bn_reduce_cl_kernel, where the last
+    tile has threads past C — regression test for the OOB read/write in
+    that tile (ADVICE.md round-1 high finding)."""
     from chainermn_amd.links import FusedBatchNorm2d
     torch.manual_seed(71)
     x32 = torch.randn(6, C, 9, 9, device=DEV)

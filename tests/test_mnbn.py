@@ -62,14 +62,13 @@ def test_mnbn_matches_global_batch_bn():
                               bn.weight.grad, atol=1e-4)
         assert torch.allclose(torch.from_numpy(r['gbeta']),
                               bn.bias.grad, atol=1e-4)
-        # running stats (torch uses unbiased var for running_var)
+        # running stats: upstream Chainer BN folds the UNBIASED batch var
+        # (m/(m-1), m = global element count) into running_var, like torch
         n = size * B * H * W
         unbias = n / (n - 1)
         batch_var = x.detach().var(dim=(0, 2, 3), unbiased=False)
-        expect_var = 0.9 * 1.0 + 0.1 * batch_var  # decay=0.9 start=1
+        expect_var = 0.9 * 1.0 + 0.1 * batch_var * unbias  # decay=0.9
         got_var = torch.from_numpy(r['running_var'])
-        # chainermn keeps biased batch var in running stats; torch uses
-        # unbiased — compare against our own definition
         assert torch.allclose(got_var, expect_var, atol=1e-5), \
             (got_var, expect_var, unbias)
 
