@@ -373,10 +373,10 @@ class FusedDataParallelAdam(FusedDataParallelSGD):
 
     def __init__(self, model, comm, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=0.0, adamw=False, overlap=True,
-                 grad_dtype=torch.bfloat16):
+                 grad_dtype=torch.bfloat16, bucket_mb=25):
         super().__init__(model, comm, lr, momentum=0.0,
                          weight_decay=weight_decay, overlap=overlap,
-                         grad_dtype=grad_dtype)
+                         grad_dtype=grad_dtype, bucket_mb=bucket_mb)
         n = self.state.plan.total
         dev = self.state.device
         self.betas = betas
@@ -416,10 +416,10 @@ class FusedDataParallelLARS(FusedDataParallelSGD):
 
     def __init__(self, model, comm, lr, momentum=0.9, weight_decay=5e-5,
                  eta=0.001, eps=1e-9, overlap=True,
-                 grad_dtype=torch.bfloat16):
+                 grad_dtype=torch.bfloat16, bucket_mb=25):
         super().__init__(model, comm, lr, momentum=momentum,
                          weight_decay=weight_decay, overlap=overlap,
-                         grad_dtype=grad_dtype)
+                         grad_dtype=grad_dtype, bucket_mb=bucket_mb)
         if self.v is None:
             n = self.state.plan.total
             self.v = torch.zeros(n, dtype=torch.float32,
