@@ -53,6 +53,9 @@ torch::Tensor bn_fwd(torch::Tensor x, torch::Tensor mean, torch::Tensor rstd,
                      torch::Tensor gamma, torch::Tensor beta);
 std::tuple<torch::Tensor, torch::Tensor> maxpool3x3s2_fwd(torch::Tensor x);
 torch::Tensor gemm_bt_bf16(torch::Tensor a, torch::Tensor b);
+torch::Tensor gemm_bt256_bf16(torch::Tensor a, torch::Tensor b);
+torch::Tensor gemm_bt256p_bf16(torch::Tensor a, torch::Tensor b,
+                               long sched);
 torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
@@ -104,4 +107,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bt_bf16", &gemm_bt_bf16,
         "hand-written MFMA bf16 GEMM: D = A @ B^T (128x128x64 tile, "
         "glds staging, XOR-swizzled LDS, XCD-aware grid)");
+  m.def("gemm_bt256_bf16", &gemm_bt256_bf16,
+        "256x256x64-tile 8-wave MFMA bf16 GEMM: D = A @ B^T");
+  m.def("gemm_bt256p_bf16", &gemm_bt256p_bf16,
+        "256-tile GEMM, 8-phase schedule (counted vmcnt, raw barriers, "
+        "setprio): D = A @ B^T.  sched bit0 = stage-before-reads, "
+        "bit1 = single barrier per phase",
+        py::arg("a"), py::arg("b"), py::arg("sched") = 0);
 }

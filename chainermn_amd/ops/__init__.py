@@ -226,6 +226,25 @@ def _time_gpu(fn, n=5):
     return (time.perf_counter() - t0) / n
 
 
+def _decide_multi(key, candidates):
+    """Pick the fastest of named candidate callables for one shape,
+    identically on every rank (see _decide)."""
+    import torch.distributed as dist
+    synced = (dist.is_available() and dist.is_initialized()
+              and dist.get_world_size() > 1)
+    if synced and dist.get_rank() != 0:
+        lst = [None]
+        dist.broadcast_object_list(lst, src=0)
+        choice = lst[0]
+    else:
+        times = {name: _time_gpu(fn) for name, fn in candidates.items()}
+        choice = min(times, key=times.get)
+        if synced:
+            dist.broadcast_object_list([choice], src=0)
+    _GEMM_CHOICE[key] = choice
+    return choice
+
+
 def _decide(key, mfma_fn, blas_fn):
     """Pick mfma-vs-blas for one shape, identically on every rank.
 
@@ -270,8 +289,13 @@ def gemm_bt(a, b):
     choice = _GEMM_CHOICE.get(key)
     if choice is None:
         ext = require()
-        choice = _decide(key, lambda: ext.gemm_bt_bf16(a, b),
-                         lambda: a @ b.t())
+        cands = {'mfma': lambda: ext.gemm_bt_bf16(a, b),
+                 'blas': lambda: a @ b.t()}
+        if a.shape[0] % 256 == 0 and b.shape[0] % 256 == 0:
+            cands['mfma256'] = lambda: ext.gemm_bt256p_bf16(a, b, 3)
+        choice = _decide_multi(key, cands)
+    if choice == 'mfma256':
+        return require().gemm_bt256p_bf16(a, b, 3)
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b)
     return a @ b.t()
@@ -290,9 +314,14 @@ def gemm_nn(a, b):
     choice = _GEMM_CHOICE.get(key)
     if choice is None:
         ext = require()
-        choice = _decide(key,
-                         lambda: ext.gemm_bt_bf16(a, b.t().contiguous()),
-                         lambda: a @ b)
+        cands = {'mfma': lambda: ext.gemm_bt_bf16(a, b.t().contiguous()),
+                 'blas': lambda: a @ b}
+        if a.shape[0] % 256 == 0 and b.shape[1] % 256 == 0:
+            cands['mfma256'] = lambda: ext.gemm_bt256p_bf16(
+                a, b.t().contiguous(), 3)
+        choice = _decide_multi(key, cands)
+    if choice == 'mfma256':
+        return require().gemm_bt256p_bf16(a, b.t().contiguous(), 3)
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b.t().contiguous())
     return a @ b

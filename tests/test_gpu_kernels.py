@@ -642,6 +642,29 @@ def test_mfma_gemm_bt_vs_matmul():
         assert err <= 0.02 * ref_mag + 0.1, (M, N, K, err, ref_mag)
 
 
+def test_mfma_gemm_bt256_vs_matmul():
+    """256x256-tile 8-wave GEMM numerics vs hipBLASLt/fp32 reference."""
+    torch.manual_seed(55)
+    for M, N, K in [(256, 256, 64), (512, 256, 128), (256, 512, 192),
+                    (1024, 512, 320), (768, 768, 64)]:
+        a = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+        b = torch.randn(N, K, device=DEV, dtype=torch.bfloat16)
+        # asymmetric structure so a row/col swap cannot pass
+        a[0, :] += torch.arange(K, device=DEV, dtype=torch.bfloat16) * 0.01
+        b[1, :] -= torch.arange(K, device=DEV, dtype=torch.bfloat16) * 0.02
+        ref = a.float() @ b.float().t()
+        ref_mag = ref.abs().max()
+        runs = [('simple', lambda: _ext().gemm_bt256_bf16(a, b))]
+        runs += [('sched%d' % s,
+                  lambda s=s: _ext().gemm_bt256p_bf16(a, b, s))
+                 for s in range(4)]
+        for name, fn in runs:
+            d = fn()
+            err = (d.float() - ref).abs().max()
+            assert err <= 0.02 * ref_mag + 0.1, \
+                (name, M, N, K, err, ref_mag)
+
+
 @pytest.mark.parametrize('C', [12, 24, 40, 1028, 2560])
 def test_bn_odd_channel_counts(C):
     """Channels-last reduce paths for C%4==0 but not power-of-two lane

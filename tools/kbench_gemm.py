@@ -11,13 +11,19 @@ def t(fn, n=30):
 
 print("bf16 GEMM D = A @ B^T, uniform random [-1,1) operands:")
 for M, N, K in [(1024,1024,1024),(4096,4096,4096),(8192,8192,8192),
-                (401408,128,512),(100352,256,1024)]:
+                (401408,128,512),(100352,256,1024),(401408,256,512),
+                (100352,512,1024),(25088,1024,2048),(6272,2048,1024)]:
     a = (torch.rand(M, K, device='cuda', dtype=torch.bfloat16)*2-1)
     b = (torch.rand(N, K, device='cuda', dtype=torch.bfloat16)*2-1)
     fl = 2.0*M*N*K
     dt1 = t(lambda: ops.require().gemm_bt_bf16(a, b))
     dt2 = t(lambda: a @ b.t())
-    print(f"  M{M} N{N} K{K}: ours {fl/dt1/1e12:7.1f} TF | hipBLASLt {fl/dt2/1e12:7.1f} TF")
+    line = f"  M{M} N{N} K{K}: 128t {fl/dt1/1e12:7.1f} TF | hipBLASLt {fl/dt2/1e12:7.1f} TF"
+    if M % 256 == 0 and N % 256 == 0:
+        dt3 = t(lambda: ops.require().gemm_bt256_bf16(a, b))
+        dt4 = t(lambda: ops.require().gemm_bt256p_bf16(a, b))
+        line += f" | 256t {fl/dt3/1e12:7.1f} TF | 256t8p {fl/dt4/1e12:7.1f} TF"
+    print(line)
 
 print("\nEXPERIMENTAL implicit-GEMM 3x3/s1/p1 conv fwd vs torch/MIOpen:")
 import torch.nn.functional as F
