@@ -1,27 +1,32 @@
-// EXPERIMENTAL: 3x3 / stride-1 / pad-1 forward convolution as implicit
-// GEMM on MFMA (gfx950), channels_last.
-//
-// Motivation (round-2 groundwork): MIOpen's asm igemm kernels for these
-// shapes use the CDNA2-era 32x32x8 MFMA — half the 32x32x16/16x16x32 rate
-// on gfx950.  This kernel reuses gemm.hip's 128x128x64 tile but builds
-// the A-operand (im2col view of the input) on the fly:
+// 3x3 / stride-1 / pad-1 convolution as implicit GEMM on gfx950 MFMA,
+// channels_last — the round-2 production kernel (fwd + dgrad; wgrad
+// stays on MIOpen's split-K wrw, see links/fused_conv.py).
 //
 //   y[m, co] = sum_{ky,kx,ci} x[n, oh+ky-1, ow+kx-1, ci] * w[co,ky,kx,ci]
 //   M = N*H*W (m -> n,oh,ow),  K = 9*Cin (k -> (ky,kx), ci)
 //
-// Cin % 64 == 0 keeps every 64-wide K-tile inside one (ky,kx) tap.  The
-// input tile is staged through registers with per-lane boundary
-// predication (padding rows/cols load zeros) and written to LDS at the
-// XOR-swizzled address directly (register staging may scatter, unlike
-// global_load_lds); the weight tile uses glds with a pre-swizzled source
-// like gemm.hip.  Single-buffered: measured 280-425 TF vs MIOpen's ~460;
-// a naive 2-buffer in-loop staging variant measured SLOWER (ordinary
-// register loads beside in-flight glds hit hipcc's vmcnt(0) drain — the
-// programming guide's mixed-load-kinds trap), so the pipelined version
-// needs the T14 split (issue loads early, ds_write after the barrier) —
-// round-2 work, see STATUS.md.
+// Structure = gemm256.hip's winning phased schedule (256-row M tile,
+// 8 waves, double-buffered LDS via global_load_lds, counted vmcnt at
+// phase ends, one raw barrier per phase, s_setprio MFMA clusters),
+// specialized per Cout:
 //
-// Not wired into any model path; exposed as ops-level experimental API.
+//   Cout >= 256: BN=256, waves 2(M)x4(N), per-wave 128x64  (16 MFMA/ph)
+//   Cout == 128: BN=128, waves 4x2,       per-wave  64x64  ( 8 MFMA/ph)
+//   Cout ==  64: BN= 64, waves 8x1,       per-wave  32x64  ( 4 MFMA/ph)
+//
+// The A operand (im2col view of x) is staged by global_load_lds with the
+// PER-LANE SOURCE address doing the im2col: each lane's (n,oh,ow) base
+// address and bounds are precomputed once per kernel (the output
+// position set of a chunk never changes across K-tiles; only the tap
+// offset does), and out-of-bounds taps redirect the lane's source to a
+// zero-filled page — no branches, no register staging, and the XOR bank
+// swizzle rides on the source address as in gemm256.hip.  Cin % 64 == 0
+// keeps every 64-wide K-tile inside one (ky,kx) tap; tap/channel
+// counters advance incrementally (no divisions in the loop).
+//
+// dgrad reuses this kernel: gx = conv3x3s1p1(gy, W^T with taps flipped)
+// — the Python side pre-transposes w once per backward (see
+// links/fused_conv.py), so both directions run the same MFMA path.
 
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
@@ -33,129 +38,276 @@ namespace {
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-#define C3_BM 128
-#define C3_BN 128
-#define C3_BK 64
-#define C3_THREADS 256
+#define C3_THREADS 512
 
 __device__ __forceinline__ int c3_swz(int row, int within_byte) {
   return within_byte ^ ((row & 7) << 4);
 }
 
-__global__ __launch_bounds__(C3_THREADS) void conv3x3s1_fwd_kernel(
-    const short* __restrict__ X,  // [N, H, W, Cin] bf16 bits
-    const short* __restrict__ Wt, // [Cout, 3, 3, Cin]
-    short* __restrict__ Y,        // [N, H, W, Cout]
-    int Nb, int H, int Wd, int Cin, int Cout) {
-  __shared__ __attribute__((aligned(16))) char lds[2 * C3_BM * C3_BK * 2];
-  char* ldsA = lds;
-  char* ldsB = lds + C3_BM * C3_BK * 2;
+template <int N>
+__device__ __forceinline__ void c3_waitbar() {
+  // N = outstanding glds ALLOWED across this barrier; N < 0: no wait.
+  if constexpr (N == 0)
+    asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+  else if constexpr (N == 1)
+    asm volatile("s_waitcnt vmcnt(1)\n\ts_barrier" ::: "memory");
+  else if constexpr (N == 2)
+    asm volatile("s_waitcnt vmcnt(2)\n\ts_barrier" ::: "memory");
+  else if constexpr (N == 3)
+    asm volatile("s_waitcnt vmcnt(3)\n\ts_barrier" ::: "memory");
+  else if constexpr (N == 4)
+    asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+  else
+    asm volatile("s_barrier" ::: "memory");
+}
 
-  const long M = (long)Nb * H * Wd;
-  const int ntile_n = Cout / C3_BN;
-  const int tile_m = blockIdx.x / ntile_n;
-  const int tile_n = blockIdx.x % ntile_n;
+// WM = per-wave output rows; WAVES_M x WAVES_N = 8 waves; BN = 64*WAVES_N.
+template <int WM, int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
+    const short* __restrict__ X,   // [N, H, W, Cin] bf16 bits
+    const short* __restrict__ Wt,  // [Cout, 3, 3, Cin]
+    const short* __restrict__ ZP,  // 128-byte zero page
+    short* __restrict__ Y,         // [N, H, W, Cout]
+    int HW, int H, int Wd, int Cin, int Cout) {
+  constexpr int BM = 256;
+  constexpr int BN = 64 * WAVES_N;
+  constexpr int MF = WM / 32;       // m-frags per quadrant
+  // stage sizes in glds instructions per thread
+  constexpr int SA = 2;                          // one A half = 16 KiB
+  constexpr int SB = (WAVES_N >= 2) ? WAVES_N / 2 : 1;  // B half (or both)
+  constexpr bool B_BOTH = (WAVES_N == 1);  // stage both B halves at p1
+  // steady-state vmcnt allowances (see scheduling note in gemm256.hip;
+  // derived from the per-config stage sizes):
+  constexpr int W_P0 = B_BOTH ? SA : (SB + SA);      // guard A1(t)
+  constexpr int W_P1 = B_BOTH ? -1 : (SA + SB);      // guard B1(t)
+  constexpr int W_P3 = B_BOTH ? SA : (SA + SB);      // guard A0/B0(t+1)
+  // peeled tile allowances
+  constexpr int P_P0 = B_BOTH ? 0 : SB;              // guard A1(last)
+  constexpr int P_P1 = B_BOTH ? -1 : 0;              // guard B1(last)
+
+  __shared__ __attribute__((aligned(16))) char lds[2 * (BM + BN) * 64 * 2];
+  auto ldsA = [&](int buf) -> char* {
+    return lds + buf * (BM + BN) * 64 * 2;
+  };
+  auto ldsB = [&](int buf) -> char* {
+    return lds + buf * (BM + BN) * 64 * 2 + BM * 64 * 2;
+  };
+
+  const int nwg = gridDim.x;
+  const int orig = blockIdx.x;
+  const int q = nwg / 8, r = nwg % 8, xcd = orig % 8;
+  const int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+                   + orig / 8;
+  const int ntile_n = Cout / BN;
+  const long tile_m = wgid / ntile_n;
+  const int tile_n = wgid % ntile_n;
+
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
-  const int wr = wave >> 1;
-  const int wc = wave & 1;
-  const int KW = 9 * Cin;
+  const int wr = wave / WAVES_N;
+  const int wc = wave % WAVES_N;
+  const int fr = lane & 15;
+  const int kq = lane >> 4;
 
-  f32x4 acc[4][4];
+  // ---- per-thread A (im2col) stage descriptors: for (h, i) the thread
+  // loads 16 B of chunk c = i*8+wave of A-half h; the output position m
+  // for that 16 B is fixed for the whole kernel ----
+  long a_base[2][2];   // X + m*Cin + swizzled element offset
+  int a_oh[2][2], a_ow[2][2];
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int h = 0; h < 2; ++h)
 #pragma unroll
-    for (int n = 0; n < 4; ++n) acc[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-
-  const int nkt = KW / C3_BK;
-  for (int kt = 0; kt < nkt; ++kt) {
-    const int k0 = kt * C3_BK;
-    const int r = k0 / Cin;          // tap index 0..8
-    const int ky = r / 3, kx = r % 3;
-    const int ci0 = k0 - r * Cin;
-    // ---- stage A: register loads with boundary predication, swizzled
-    // ds_write (2 chunks of 16 B per thread cover 128x64 bf16) ----
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int p = (i * C3_THREADS + t) * 16;
-      const int row = p >> 7;
+    for (int i = 0; i < 2; ++i) {
+      const int c = i * 8 + wave;
+      constexpr int CPW = WM / 16;  // chunks per wave-row block per half
+      const int row0 = (c / CPW) * WM + h * (WM / 2) + (c % CPW) * 8;
+      const int p = lane * 16;
+      const int row = row0 + (p >> 7);
       const int within = p & 127;
-      const long m = (long)tile_m * C3_BM + row;
-      uint4 v = {0u, 0u, 0u, 0u};
-      if (m < M) {
-        const long n = m / ((long)H * Wd);
-        const long rem = m - n * (long)H * Wd;
-        const int oh = (int)(rem / Wd);
-        const int ow = (int)(rem - (long)oh * Wd);
-        const int ih = oh + ky - 1;
-        const int iw = ow + kx - 1;
-        if (ih >= 0 && ih < H && iw >= 0 && iw < Wd) {
-          const short* src = X + (((n * H + ih) * (long)Wd + iw) * Cin
-                                  + ci0 + (within >> 1));
-          v = *reinterpret_cast<const uint4*>(src);
-        }
-      }
-      *reinterpret_cast<uint4*>(ldsA + row * 128 + c3_swz(row, within)) = v;
+      const long m = tile_m * BM + row;
+      const unsigned mu = (unsigned)m;
+      const unsigned n = mu / (unsigned)HW;
+      const unsigned rem = mu - n * (unsigned)HW;
+      const int oh = (int)(rem / (unsigned)Wd);
+      const int ow = (int)(rem - (unsigned)oh * (unsigned)Wd);
+      a_oh[h][i] = oh;
+      a_ow[h][i] = ow;
+      a_base[h][i] = m * Cin + (c3_swz(row, within) >> 1);
     }
-    // ---- stage B: weights [Cout][9][Cin]; row = co within tile ----
+
+  // stage A-half h of K-tile at (tap ky,kx, channel block ci0)
+  auto stage_a = [&](int buf, int h, int ky, int kx, int ci0) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      // glds needs a WAVE-UNIFORM LDS base (+ lane*16 implicit)
-      const int base = (i * C3_THREADS + wave * 64) * 16;
-      const int p = base + lane * 16;
-      const int row = p >> 7;
+    for (int i = 0; i < 2; ++i) {
+      const int c = i * 8 + wave;
+      constexpr int CPW = WM / 16;
+      const int row0 = (c / CPW) * WM + h * (WM / 2) + (c % CPW) * 8;
+      const int ih = a_oh[h][i] + ky - 1;
+      const int iw = a_ow[h][i] + kx - 1;
+      const bool ok = (unsigned)ih < (unsigned)H
+                      && (unsigned)iw < (unsigned)Wd;
+      const long off = a_base[h][i]
+                       + (long)((ky - 1) * Wd + (kx - 1)) * Cin + ci0;
+      const short* src = ok ? X + off : ZP;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              (ldsA(buf) + row0 * 128),
+          16, 0, 0);
+    }
+  };
+
+  // stage B (weights): half h (or both for WAVES_N==1); k0 = global k
+  auto stage_b = [&](int buf, int h, int k0) {
+    const int KW = 9 * Cin;
+#pragma unroll
+    for (int i = 0; i < SB; ++i) {
+      int row0;
+      if constexpr (B_BOTH) {
+        row0 = wave * 8;
+      } else {
+        const int c = i * 8 + wave;  // SB*8 chunks over 8 waves
+        row0 = (c / 4) * 64 + h * 32 + (c & 3) * 8;
+      }
+      const int p = lane * 16;
+      const int row = row0 + (p >> 7);
       const int within = p & 127;
-      const short* src = Wt + ((long)(tile_n * C3_BN + row)) * KW + k0
+      const short* src = Wt + ((long)(tile_n * BN + row)) * KW + k0
                          + (c3_swz(row, within) >> 1);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)
-              (ldsB + base),
+              (ldsB(buf) + row0 * 128),
           16, 0, 0);
     }
-    __syncthreads();
-    const int fr = lane & 15;
-    const int kq = lane >> 4;
+  };
+
+  f32x4 acc[2 * MF][4];
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      bf16x8 a_frag[4], b_frag[4];
+  for (int m = 0; m < 2 * MF; ++m)
 #pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        const int row = wr * 64 + m * 16 + fr;
-        const int kbyte = ks * 64 + kq * 16;
-        a_frag[m] = *(const bf16x8*)(ldsA + row * 128 + c3_swz(row, kbyte));
-      }
-#pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        const int row = wc * 64 + n * 16 + fr;
-        const int kbyte = ks * 64 + kq * 16;
-        b_frag[n] = *(const bf16x8*)(ldsB + row * 128 + c3_swz(row, kbyte));
-      }
-#pragma unroll
-      for (int m = 0; m < 4; ++m)
-#pragma unroll
-        for (int n = 0; n < 4; ++n)
-          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    for (int n = 0; n < 4; ++n) acc[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 af[2][MF];
+  bf16x8 bfr[2][2];
+
+#define C3_READ_A(buf, qa)                                                   \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+      _Pragma("unroll") for (int mi = 0; mi < MF; ++mi) {                    \
+    const int row = wr * WM + (qa) * (WM / 2) + mi * 16 + fr;                \
+    af[ks][mi] = *(const bf16x8*)(ldsA(buf) + row * 128                      \
+                                  + c3_swz(row, ks * 64 + kq * 16));         \
+  }
+#define C3_READ_B(buf, qb)                                                   \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+      _Pragma("unroll") for (int ni = 0; ni < 2; ++ni) {                     \
+    const int row = wc * 64 + (qb) * 32 + ni * 16 + fr;                      \
+    bfr[ks][ni] = *(const bf16x8*)(ldsB(buf) + row * 128                     \
+                                   + c3_swz(row, ks * 64 + kq * 16));        \
+  }
+#define C3_MFMA(qa, qb)                                                      \
+  __builtin_amdgcn_s_setprio(1);                                            \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+      _Pragma("unroll") for (int mi = 0; mi < MF; ++mi)                      \
+          _Pragma("unroll") for (int ni = 0; ni < 2; ++ni)                   \
+    acc[(qa) * MF + mi][(qb) * 2 + ni] =                                     \
+        __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             \
+            af[ks][mi], bfr[ks][ni], acc[(qa) * MF + mi][(qb) * 2 + ni],     \
+            0, 0, 0);                                                        \
+  __builtin_amdgcn_s_setprio(0);
+#define C3_BAR() asm volatile("s_barrier" ::: "memory");
+
+  const int nkt = 9 * (Cin / 64);
+  // tap/channel counters for the tile being staged (one ahead); the
+  // current tile's taps are never needed (its data is already in LDS)
+  int sky = 0, skx = 0, sci0 = 0, sk0 = 0;
+  auto advance = [&](int& ky_, int& kx_, int& ci0_) {
+    ci0_ += 64;
+    if (ci0_ == Cin) {
+      ci0_ = 0;
+      if (++kx_ == 3) { kx_ = 0; ++ky_; }
     }
-    __syncthreads();
+  };
+
+  // prologue: all halves of tile 0 in stage order (A0, B0/B01, A1, B1);
+  // A0 AND B0 must land (for B_BOTH the B stage is 3rd-from-newest, so
+  // the allowance is SA, not SA+SB)
+  stage_a(0, 0, 0, 0, 0);
+  stage_b(0, 0, 0);
+  stage_a(0, 1, 0, 0, 0);
+  if constexpr (!B_BOTH) stage_b(0, 1, 0);
+  c3_waitbar<B_BOTH ? SA : SA + SB>();
+
+  for (int kt = 0; kt < nkt - 1; ++kt) {
+    const int buf = kt & 1;
+    const int nxt = buf ^ 1;
+    advance(sky, skx, sci0);
+    sk0 += 64;
+    // phase 0: quadrant (0,0); stage A-half0(kt+1)
+    stage_a(nxt, 0, sky, skx, sci0);
+    C3_READ_A(buf, 0)
+    C3_READ_B(buf, 0)
+    c3_waitbar<W_P0>();
+    C3_MFMA(0, 0)
+    // phase 1: quadrant (1,0); stage B-half0 (or both halves)
+    stage_b(nxt, 0, sk0);
+    C3_READ_A(buf, 1)
+    c3_waitbar<W_P1>();
+    C3_MFMA(1, 0)
+    // phase 2: quadrant (1,1); stage A-half1
+    stage_a(nxt, 1, sky, skx, sci0);
+    C3_READ_B(buf, 1)
+    C3_BAR()
+    C3_MFMA(1, 1)
+    // phase 3: quadrant (0,1); stage B-half1 (unless staged at p1)
+    if constexpr (!B_BOTH) stage_b(nxt, 1, sk0);
+    C3_READ_A(buf, 0)
+    c3_waitbar<W_P3>();
+    C3_MFMA(0, 1)
   }
 
-  const int col0 = tile_n * C3_BN + wc * 64 + (lane & 15);
-  const long row_base = (long)tile_m * C3_BM + wr * 64 + (lane >> 4) * 4;
+  {  // peeled last tile
+    const int buf = (nkt - 1) & 1;
+    C3_READ_A(buf, 0)
+    C3_READ_B(buf, 0)
+    c3_waitbar<P_P0>();
+    C3_MFMA(0, 0)
+    C3_READ_A(buf, 1)
+    c3_waitbar<P_P1>();
+    C3_MFMA(1, 0)
+    C3_READ_B(buf, 1)
+    C3_BAR()
+    C3_MFMA(1, 1)
+    C3_READ_A(buf, 0)
+    C3_MFMA(0, 1)
+  }
+
+#undef C3_READ_A
+#undef C3_READ_B
+#undef C3_MFMA
+#undef C3_BAR
+
+  const int col0 = tile_n * BN + wc * 64 + fr;
+  const long row_base = tile_m * BM + wr * WM + (lane >> 4) * 4;
 #pragma unroll
-  for (int m = 0; m < 4; ++m) {
+  for (int m = 0; m < 2 * MF; ++m) {
 #pragma unroll
     for (int rg = 0; rg < 4; ++rg) {
       const long row = row_base + m * 16 + rg;
-      if (row >= M) continue;
 #pragma unroll
       for (int n = 0; n < 4; ++n)
         Y[row * Cout + col0 + n * 16] =
             (short)c10::BFloat16(acc[m][n][rg]).x;
     }
   }
+}
+
+torch::Tensor& zero_page(const torch::Tensor& like) {
+  static torch::Tensor zp;
+  if (!zp.defined() || zp.device() != like.device())
+    zp = torch::zeros({64}, like.options().dtype(at::kBFloat16));
+  return zp;
 }
 
 }  // namespace
@@ -172,18 +324,38 @@ torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w) {
   const long Nb = x.size(0);
   const int Cin = (int)x.size(1), H = (int)x.size(2), Wd = (int)x.size(3);
   const int Cout = (int)w.size(0);
-  TORCH_CHECK(Cin % 64 == 0 && Cout % C3_BN == 0,
-              "requires Cin%64==0 and Cout%128==0");
-  const long M = Nb * H * Wd;
-  TORCH_CHECK(M % C3_BM == 0, "requires N*H*W % 128 == 0");
+  TORCH_CHECK(Cin % 64 == 0 && Cout % 64 == 0,
+              "requires Cin%64==0 and Cout%64==0");
+  const long M = Nb * (long)H * Wd;
+  TORCH_CHECK(M % 256 == 0, "requires N*H*W % 256 == 0");
+  TORCH_CHECK(M < (1L << 31), "N*H*W must fit in 32 bits");
   auto y = torch::empty({Nb, Cout, H, Wd},
                         x.options().memory_format(
                             at::MemoryFormat::ChannelsLast));
   auto stream = c10::hip::getCurrentHIPStream();
-  const long grid = (M / C3_BM) * (Cout / C3_BN);
-  hipLaunchKernelGGL(conv3x3s1_fwd_kernel, dim3((unsigned)grid),
-                     dim3(C3_THREADS), 0, stream.stream(),
-                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
-                     (short*)y.data_ptr(), (int)Nb, H, Wd, Cin, Cout);
+  const short* xp = (const short*)x.data_ptr();
+  const short* wp = (const short*)w.data_ptr();
+  const short* zp = (const short*)zero_page(x).data_ptr();
+  short* yp = (short*)y.data_ptr();
+  const int HW = H * Wd;
+  if (Cout % 256 == 0) {
+    const long grid = (M / 256) * (Cout / 256);
+    hipLaunchKernelGGL((conv3x3_igemm_kernel<128, 2, 4>),
+                       dim3((unsigned)grid), dim3(C3_THREADS), 0,
+                       stream.stream(), xp, wp, zp, yp, HW, H, Wd, Cin,
+                       Cout);
+  } else if (Cout % 128 == 0) {
+    const long grid = (M / 256) * (Cout / 128);
+    hipLaunchKernelGGL((conv3x3_igemm_kernel<64, 4, 2>),
+                       dim3((unsigned)grid), dim3(C3_THREADS), 0,
+                       stream.stream(), xp, wp, zp, yp, HW, H, Wd, Cin,
+                       Cout);
+  } else {
+    const long grid = (M / 256) * (Cout / 64);
+    hipLaunchKernelGGL((conv3x3_igemm_kernel<32, 8, 1>),
+                       dim3((unsigned)grid), dim3(C3_THREADS), 0,
+                       stream.stream(), xp, wp, zp, yp, HW, H, Wd, Cin,
+                       Cout);
+  }
   return y;
 }

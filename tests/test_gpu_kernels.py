@@ -714,10 +714,13 @@ def test_pure_rccl_mean_grad_async_world1():
 
 
 def test_conv3x3_implicit_gemm_fwd():
-    """EXPERIMENTAL implicit-GEMM 3x3 conv forward vs F.conv2d."""
+    """Implicit-GEMM 3x3/s1/p1 conv forward vs F.conv2d, all three wave
+    geometries (Cout 64 / 128 / 256) incl. border-dominated shapes."""
     torch.manual_seed(81)
     for N, C, H, W, Cout in [(2, 64, 16, 16, 128), (1, 128, 32, 32, 128),
-                             (2, 64, 32, 16, 256)]:
+                             (2, 64, 32, 16, 256), (16, 64, 4, 4, 64),
+                             (4, 128, 8, 8, 64), (1, 192, 16, 16, 320),
+                             (16, 128, 28, 28, 128)]:
         x = torch.randn(N, C, H, W, device=DEV, dtype=torch.bfloat16
                         ).to(memory_format=torch.channels_last)
         w = torch.randn(Cout, C, 3, 3, device=DEV, dtype=torch.bfloat16
