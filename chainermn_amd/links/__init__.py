@@ -7,4 +7,4 @@ from chainermn_amd.links.fused_batch_norm import FusedBatchNorm2d  # noqa: F401
 from chainermn_amd.links.fused_batch_norm import FusedBNReLU2d  # noqa: F401
 from chainermn_amd.links.batch_normalization import MultiNodeBNReLU2d  # noqa: F401
 from chainermn_amd.links.fused_pool import FusedMaxPool2d  # noqa: F401
-from chainermn_amd.links.fused_conv import FusedConv1x1  # noqa: F401
+from chainermn_amd.links.fused_conv import FusedConv1x1, FusedConv3x3  # noqa: F401

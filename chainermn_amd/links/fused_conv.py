@@ -83,3 +83,59 @@ class FusedConv1x1(nn.Module):
     def extra_repr(self):
         return '{}, {}, kernel_size=1 (GEMM)'.format(self.in_channels,
                                                      self.out_channels)
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """3x3/s1/p1 conv on the phased implicit-GEMM MFMA kernel.
+
+    dgrad REUSES the forward kernel: gx = conv3x3s1p1(gy, W^T with taps
+    flipped) — the transpose/flip is a 4.7 MB-max copy per backward,
+    trivial next to the conv.  wgrad stays on MIOpen's split-K wrw
+    kernels (same reasoning as _Conv1x1Fn)."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        return ops.conv3x3(x, w)
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        gy = gy.contiguous(memory_format=torch.channels_last)
+        wt = w.flip(2, 3).permute(1, 0, 2, 3).contiguous(
+            memory_format=torch.channels_last)
+        gx = ops.conv3x3(gy, wt)
+        _, gw, _ = torch.ops.aten.convolution_backward(
+            gy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [False, True, False])
+        return gx, gw
+
+
+class FusedConv3x3(nn.Module):
+    """Drop-in for nn.Conv2d(cin, cout, 3, stride, padding=1, bias=False).
+
+    stride-1 channels_last bf16 inputs run the hand-written CDNA4
+    implicit-GEMM kernel (autotuned vs MIOpen per shape); anything else
+    (stride 2, CPU, fp32) falls back to F.conv2d."""
+
+    def __init__(self, in_channels, out_channels, stride=1):
+        super().__init__()
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        self.stride = stride
+        ref = nn.Conv2d(in_channels, out_channels, 3, stride=stride,
+                        padding=1, bias=False)
+        self.weight = ref.weight
+
+    def forward(self, x):
+        if (self.stride == 1 and x.is_cuda and x.dim() == 4
+                and x.dtype == torch.bfloat16
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and not x.is_contiguous()):
+            return _Conv3x3Fn.apply(x, self.weight)
+        return torch.nn.functional.conv2d(x, self.weight,
+                                          stride=self.stride, padding=1)
+
+    def extra_repr(self):
+        return '{}, {}, kernel_size=3, stride={} (implicit GEMM)'.format(
+            self.in_channels, self.out_channels, self.stride)

@@ -23,15 +23,18 @@ class Bottleneck(nn.Module):
         self.fused_ops = fused_ops
         if fused_ops:
             bn_relu, bn_plain = _fused_bn_classes(comm)
-            from chainermn_amd.links.fused_conv import FusedConv1x1
+            from chainermn_amd.links.fused_conv import (FusedConv1x1,
+                                                        FusedConv3x3)
             conv1x1 = FusedConv1x1
+            conv3x3 = FusedConv3x3
         else:
             bn_relu = bn_plain = norm_layer
             conv1x1 = (lambda ci, co: nn.Conv2d(ci, co, 1, bias=False))
+            conv3x3 = (lambda ci, co, stride=1: nn.Conv2d(
+                ci, co, 3, stride=stride, padding=1, bias=False))
         self.conv1 = conv1x1(in_ch, width)
         self.bn1 = bn_relu(width)
-        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
-                               bias=False)
+        self.conv2 = conv3x3(width, width, stride=stride)
         self.bn2 = bn_relu(width)
         self.conv3 = conv1x1(width, out_ch)
         self.bn3 = bn_plain(out_ch)
@@ -98,9 +101,10 @@ class ResNet50(nn.Module):
         self.layer4 = self._make_layer(512, 3, stride=2)
         self.avgpool = nn.AdaptiveAvgPool2d(1)
         self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
-        from chainermn_amd.links.fused_conv import FusedConv1x1
+        from chainermn_amd.links.fused_conv import (FusedConv1x1,
+                                                    FusedConv3x3)
         for m in self.modules():
-            if isinstance(m, (nn.Conv2d, FusedConv1x1)):
+            if isinstance(m, (nn.Conv2d, FusedConv1x1, FusedConv3x3)):
                 nn.init.kaiming_normal_(m.weight, mode='fan_out',
                                         nonlinearity='relu')
             elif isinstance(m, nn.BatchNorm2d):

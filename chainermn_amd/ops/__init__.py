@@ -325,3 +325,29 @@ def gemm_nn(a, b):
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b.t().contiguous())
     return a @ b
+
+
+def conv3x3(x, w):
+    """3x3/stride-1/pad-1 convolution, channels_last bf16: per-shape
+    autotune between the phased implicit-GEMM MFMA kernel (conv3x3.hip —
+    1.55-2.0x MIOpen on ResNet-50's 3x3 shapes, profiles/conv3x3_v2.txt)
+    and MIOpen via F.conv2d."""
+    import torch.nn.functional as F
+    eligible = (x.is_cuda and x.dtype == torch.bfloat16
+                and w.dtype == torch.bfloat16 and x.dim() == 4
+                and x.shape[1] % 64 == 0 and w.shape[0] % 64 == 0
+                and (x.shape[0] * x.shape[2] * x.shape[3]) % 256 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and w.is_contiguous(memory_format=torch.channels_last))
+    if not eligible:
+        return F.conv2d(x, w, padding=1)
+    key = ('c3', tuple(x.shape), w.shape[0])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        ext = require()
+        choice = _decide_multi(key, {
+            'mfma': lambda: ext.conv3x3s1_fwd_bf16(x, w),
+            'miopen': lambda: F.conv2d(x, w, padding=1)})
+    if choice == 'mfma':
+        return require().conv3x3s1_fwd_bf16(x, w)
+    return F.conv2d(x, w, padding=1)

@@ -730,3 +730,48 @@ def test_conv3x3_implicit_gemm_fwd():
         err = (y.float() - ref.float()).abs().max()
         mag = ref.float().abs().max()
         assert err <= 0.03 * mag + 0.05, (N, C, H, W, Cout, err, mag)
+
+
+def test_fused_conv3x3_fwd_bwd_vs_conv2d():
+    """FusedConv3x3 autograd (custom fwd + dgrad-via-flipped-transpose +
+    MIOpen wgrad) vs F.conv2d, bf16 channels_last."""
+    from chainermn_amd.links import FusedConv3x3
+    torch.manual_seed(52)
+    for N, C, H, W, Cout in [(4, 64, 8, 8, 64), (2, 128, 16, 16, 128),
+                             (1, 64, 16, 16, 256)]:
+        x32 = torch.randn(N, C, H, W, device=DEV)
+        gy32 = torch.randn(N, Cout, H, W, device=DEV) * 0.1
+        conv = FusedConv3x3(C, Cout).to(DEV, torch.bfloat16)
+        with torch.no_grad():
+            conv.weight.mul_(0.05)
+        conv = conv.to(memory_format=torch.channels_last)
+        x = x32.to(torch.bfloat16).to(memory_format=torch.channels_last)
+        x.requires_grad_(True)
+        xr = x32.to(torch.bfloat16).clone().requires_grad_(True)
+        wr = conv.weight.detach().clone().requires_grad_(True)
+        y = conv(x)
+        yr = torch.nn.functional.conv2d(xr, wr, padding=1)
+        mag = yr.float().abs().max()
+        assert (y.float() - yr.float()).abs().max() <= 0.03 * mag + 0.05, \
+            (N, C, H, W, Cout)
+        gy = gy32.to(torch.bfloat16)
+        y.backward(gy.to(memory_format=torch.channels_last))
+        yr.backward(gy)
+        gmag = xr.grad.float().abs().max()
+        assert (x.grad.float() - xr.grad.float()).abs().max() \
+            <= 0.05 * gmag + 0.05, (N, C, H, W, Cout, 'dgrad')
+        wmag = wr.grad.float().abs().max()
+        assert (conv.weight.grad.float() - wr.grad.float()).abs().max() \
+            <= 0.05 * wmag + 0.05, (N, C, H, W, Cout, 'wgrad')
+
+
+def test_fused_conv3x3_stride2_fallback():
+    """stride-2 FusedConv3x3 must match F.conv2d exactly (fallback)."""
+    from chainermn_amd.links import FusedConv3x3
+    torch.manual_seed(53)
+    conv = FusedConv3x3(64, 128, stride=2).to(DEV, torch.bfloat16)
+    x = torch.randn(2, 64, 16, 16, device=DEV, dtype=torch.bfloat16
+                    ).to(memory_format=torch.channels_last)
+    y = conv(x)
+    yr = torch.nn.functional.conv2d(x, conv.weight, stride=2, padding=1)
+    assert torch.equal(y.float(), yr.float())
