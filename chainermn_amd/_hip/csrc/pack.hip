@@ -56,6 +56,14 @@ __global__ void batched_copy_kernel(const long* __restrict__ ptrs,
     const long local = base - seg_start;
     float r[8];
     if (PACK) {
+      if (psrc == nullptr) {
+        // K8 zero-fill: params that produced no gradient this step pack
+        // as zeros (keeps the flat layout rank-uniform)
+#pragma unroll
+        for (int k = 0; k < 8; ++k) r[k] = 0.f;
+        V8<DST>::store(flat_dst + base, r);
+        continue;
+      }
       if (local + 8 <= size) {
         V8<SRC>::load(psrc + local, r);
       } else {

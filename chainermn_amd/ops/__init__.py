@@ -72,8 +72,9 @@ class PackPlan(object):
         self.sizes = torch.tensor(sizes, dtype=torch.long, device=device)
 
     def grad_ptrs(self):
-        ptrs = torch.tensor([p.grad.data_ptr() for p in self.params],
-                            dtype=torch.long)
+        # missing grads pack as zeros (ptr 0 -> kernel zero-fill, K8)
+        ptrs = torch.tensor([0 if p.grad is None else p.grad.data_ptr()
+                             for p in self.params], dtype=torch.long)
         return ptrs.to(self.device, non_blocking=True)
 
     def data_ptrs(self):
@@ -83,14 +84,20 @@ class PackPlan(object):
 
 
 def pack_grads(plan, out_flat):
-    """Gather every param.grad into ``out_flat`` (casting to its dtype)."""
+    """Gather every param.grad into ``out_flat`` (casting to its dtype).
+    Params whose grad is None pack as zeros (K8 zero-fill)."""
     if out_flat.is_cuda:
+        src_dtype = next((p.grad.dtype for p in plan.params
+                          if p.grad is not None),
+                         plan.params[0].data.dtype)
         require().batched_pack(plan.grad_ptrs(), plan.sizes, plan.offsets,
-                               out_flat, plan.params[0].grad.dtype)
+                               out_flat, src_dtype)
         return
     out_flat.zero_()
     for p, off, n in zip(plan.params, plan.offsets_cpu, plan.sizes_cpu):
-        out_flat[off:off + n].copy_(p.grad.reshape(-1).to(out_flat.dtype))
+        if p.grad is not None:
+            out_flat[off:off + n].copy_(
+                p.grad.reshape(-1).to(out_flat.dtype))
 
 
 def unpack_grads(plan, in_flat, scale=1.0):
