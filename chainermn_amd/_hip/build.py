@@ -33,6 +33,13 @@ def _sources():
     return srcs
 
 
+def _digest_inputs():
+    # headers participate in the staleness digest (but not compilation)
+    return _sources() + [os.path.join(CSRC, f)
+                         for f in sorted(os.listdir(CSRC))
+                         if f.endswith('.h')]
+
+
 def _digest(srcs):
     h = hashlib.sha256()
     for s in srcs:
@@ -44,7 +51,7 @@ def _digest(srcs):
 def build(force=False, verbose=True):
     import torch
     srcs = _sources()
-    digest = _digest(srcs)
+    digest = _digest(_digest_inputs())
     if not force and os.path.exists(OUT) and os.path.exists(STAMP):
         with open(STAMP) as f:
             if f.read().strip() == digest:

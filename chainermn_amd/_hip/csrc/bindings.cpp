@@ -56,7 +56,10 @@ torch::Tensor gemm_bt_bf16(torch::Tensor a, torch::Tensor b);
 torch::Tensor gemm_bt256_bf16(torch::Tensor a, torch::Tensor b);
 torch::Tensor gemm_bt256p_bf16(torch::Tensor a, torch::Tensor b,
                                long sched);
-torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w);
+torch::Tensor gemm_bt256p_stats_bf16(torch::Tensor a, torch::Tensor b,
+                                     torch::Tensor stats, long sched);
+torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                                 torch::Tensor stats);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -102,8 +105,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "channels-last 3x3/s2/p1 maxpool forward (+window codes)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd,
         "gather-based maxpool backward (no atomics)");
-  m.def("conv3x3s1_fwd_bf16", &conv3x3s1_fwd_bf16,
-        "EXPERIMENTAL implicit-GEMM 3x3/s1/p1 conv forward (MFMA)");
+  m.def("conv3x3s1_fwd_bf16",
+        [](torch::Tensor x, torch::Tensor w,
+           c10::optional<torch::Tensor> stats) {
+          return conv3x3s1_fwd_bf16(
+              x, w, stats.has_value() ? *stats : torch::Tensor());
+        },
+        "implicit-GEMM 3x3/s1/p1 conv forward (phased MFMA); optional "
+        "fused BN-stats partials [16, 2*Cout]",
+        py::arg("x"), py::arg("w"), py::arg("stats") = py::none());
+  m.def("gemm_bt256p_stats_bf16", &gemm_bt256p_stats_bf16,
+        "256-tile phased GEMM with fused per-column BN-stats partials",
+        py::arg("a"), py::arg("b"), py::arg("stats"),
+        py::arg("sched") = 3);
   m.def("gemm_bt_bf16", &gemm_bt_bf16,
         "hand-written MFMA bf16 GEMM: D = A @ B^T (128x128x64 tile, "
         "glds staging, XOR-swizzled LDS, XCD-aware grid)");

@@ -130,3 +130,63 @@ __device__ __forceinline__ int find_segment(const long* offsets, int n,
   }
   return lo;
 }
+
+#define CMN_STATS_BUCKETS 16
+
+// BN-stats epilogue (shared by conv3x3 + gemm256): per-channel
+// (sum y, sum y^2) of the block's bf16-rounded output, reduced
+// wave -> block (LDS) and atomically added into SP[bucket][2*Cout]
+// (CMN_STATS_BUCKETS buckets hashed by blockIdx bound the per-address
+// atomic chains; the Python side sums the 16 rows).  Saves the
+// standalone bn_reduce read pass over the activation for the BN that
+// consumes this output.  Requires a barrier before (LDS reuse) — the
+// caller provides it.  (Defined here so conv3x3.hip and gemm256.hip share it.)
+template <int MF, int WAVES_M, int WAVES_N, typename ACC>
+__device__ __forceinline__ void stats_epilogue(
+    float* __restrict__ SP, char* lds, const ACC& acc, int lane, int wave,
+    int wr, int wc, int tile_n, int Cout, unsigned bid) {
+  constexpr int BN = 64 * WAVES_N;
+  float s[4], qq[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) { s[n] = 0.f; qq[n] = 0.f; }
+#pragma unroll
+  for (int m = 0; m < 2 * MF; ++m)
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const float v = float(c10::BFloat16(acc[m][n][rg]));
+        s[n] += v;
+        qq[n] += v * v;
+      }
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    s[n] += __shfl_xor(s[n], 16);
+    s[n] += __shfl_xor(s[n], 32);
+    qq[n] += __shfl_xor(qq[n], 16);
+    qq[n] += __shfl_xor(qq[n], 32);
+  }
+  float* sl = reinterpret_cast<float*>(lds);  // [wave][2][64]
+  if (lane < 16) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      sl[wave * 128 + lane + n * 16] = s[n];
+      sl[wave * 128 + 64 + lane + n * 16] = qq[n];
+    }
+  }
+  __syncthreads();
+  if (wr == 0) {  // one wave per distinct 64-col set; one lane per col
+    float ss = 0.f, qs = 0.f;
+#pragma unroll
+    for (int k = 0; k < WAVES_M; ++k) {
+      const int w2 = k * WAVES_N + wc;
+      ss += sl[w2 * 128 + lane];
+      qs += sl[w2 * 128 + 64 + lane];
+    }
+    const int c = tile_n * BN + wc * 64 + lane;
+    const long base = (long)(bid & (CMN_STATS_BUCKETS - 1)) * 2 * Cout;
+    atomicAdd(&SP[base + c], ss);
+    atomicAdd(&SP[base + Cout + c], qs);
+  }
+}
+

@@ -68,6 +68,7 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
     const short* __restrict__ Wt,  // [Cout, 3, 3, Cin]
     const short* __restrict__ ZP,  // 128-byte zero page
     short* __restrict__ Y,         // [N, H, W, Cout]
+    float* __restrict__ SP,        // BN stats partials or nullptr
     int HW, int H, int Wd, int Cin, int Cout) {
   constexpr int BM = 256;
   constexpr int BN = 64 * WAVES_N;
@@ -301,6 +302,12 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
             (short)c10::BFloat16(acc[m][n][rg]).x;
     }
   }
+  if (SP != nullptr) {
+    __syncthreads();  // all waves done with the main LDS buffers
+    stats_epilogue<MF, WAVES_M, WAVES_N>(SP, lds, acc, lane, wave, wr,
+                                         wc, tile_n, Cout,
+                                         (unsigned)blockIdx.x);
+  }
 }
 
 torch::Tensor& zero_page(const torch::Tensor& like) {
@@ -312,7 +319,8 @@ torch::Tensor& zero_page(const torch::Tensor& like) {
 
 }  // namespace
 
-torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w) {
+torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                                 torch::Tensor stats) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16
               && w.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.dim() == 4
@@ -337,25 +345,33 @@ torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w) {
   const short* wp = (const short*)w.data_ptr();
   const short* zp = (const short*)zero_page(x).data_ptr();
   short* yp = (short*)y.data_ptr();
+  float* sp = nullptr;
+  if (stats.defined() && stats.numel() > 0) {
+    TORCH_CHECK(stats.scalar_type() == at::kFloat && stats.is_cuda()
+                && stats.numel() == 16 * 2 * (long)Cout
+                && stats.is_contiguous(),
+                "stats must be zeroed float32 [16, 2*Cout]");
+    sp = stats.data_ptr<float>();
+  }
   const int HW = H * Wd;
   if (Cout % 256 == 0) {
     const long grid = (M / 256) * (Cout / 256);
     hipLaunchKernelGGL((conv3x3_igemm_kernel<128, 2, 4>),
                        dim3((unsigned)grid), dim3(C3_THREADS), 0,
-                       stream.stream(), xp, wp, zp, yp, HW, H, Wd, Cin,
-                       Cout);
+                       stream.stream(), xp, wp, zp, yp, sp, HW, H, Wd,
+                       Cin, Cout);
   } else if (Cout % 128 == 0) {
     const long grid = (M / 256) * (Cout / 128);
     hipLaunchKernelGGL((conv3x3_igemm_kernel<64, 4, 2>),
                        dim3((unsigned)grid), dim3(C3_THREADS), 0,
-                       stream.stream(), xp, wp, zp, yp, HW, H, Wd, Cin,
-                       Cout);
+                       stream.stream(), xp, wp, zp, yp, sp, HW, H, Wd,
+                       Cin, Cout);
   } else {
     const long grid = (M / 256) * (Cout / 64);
     hipLaunchKernelGGL((conv3x3_igemm_kernel<32, 8, 1>),
                        dim3((unsigned)grid), dim3(C3_THREADS), 0,
-                       stream.stream(), xp, wp, zp, yp, HW, H, Wd, Cin,
-                       Cout);
+                       stream.stream(), xp, wp, zp, yp, sp, HW, H, Wd,
+                       Cin, Cout);
   }
   return y;
 }

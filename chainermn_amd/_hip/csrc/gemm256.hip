@@ -170,7 +170,7 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256_bf16_kernel(
 template <int SCHED>
 __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
-    short* __restrict__ D, int M, int N, int K) {
+    short* __restrict__ D, float* __restrict__ SP, int M, int N, int K) {
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * G2_BM * G2_BK * 2];
   auto ldsA = [&](int buf) -> char* {
     return lds + buf * 2 * G2_BM * G2_BK * 2;
@@ -364,12 +364,18 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
       }
     }
   }
+  if (SP != nullptr) {
+    __syncthreads();
+    stats_epilogue<4, 2, 4>(SP, lds, acc, lane, wave, wr, wc, tile_n, N,
+                            (unsigned)blockIdx.x);
+  }
 }
 
 }  // namespace
 
 static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
-                                    int phased_sched) {
+                                    int phased_sched,
+                                    torch::Tensor stats = {}) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16
               && b.scalar_type() == at::kBFloat16);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
@@ -383,6 +389,15 @@ static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
   const short* ap = (const short*)a.data_ptr();
   const short* bp = (const short*)b.data_ptr();
   short* dp = (short*)d.data_ptr();
+  float* sp = nullptr;
+  if (stats.defined() && stats.numel() > 0) {
+    TORCH_CHECK(phased_sched > 0, "stats need the phased kernel");
+    TORCH_CHECK(stats.scalar_type() == at::kFloat && stats.is_cuda()
+                && stats.numel() == 16 * 2 * (long)N
+                && stats.is_contiguous(),
+                "stats must be zeroed float32 [16, 2*N]");
+    sp = stats.data_ptr<float>();
+  }
   switch (phased_sched) {
     case 0:
       hipLaunchKernelGGL(gemm_bt256_bf16_kernel, dim3(grid),
@@ -392,22 +407,22 @@ static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
     case 1:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<0>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         M, N, K);
+                         sp, M, N, K);
       break;
     case 2:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<1>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         M, N, K);
+                         sp, M, N, K);
       break;
     case 3:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<2>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         M, N, K);
+                         sp, M, N, K);
       break;
     default:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<3>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         M, N, K);
+                         sp, M, N, K);
   }
   return d;
 }
@@ -419,4 +434,9 @@ torch::Tensor gemm_bt256_bf16(torch::Tensor a, torch::Tensor b) {
 torch::Tensor gemm_bt256p_bf16(torch::Tensor a, torch::Tensor b,
                                long sched) {
   return launch_gemm256(a, b, 1 + (int)sched);
+}
+
+torch::Tensor gemm_bt256p_stats_bf16(torch::Tensor a, torch::Tensor b,
+                                     torch::Tensor stats, long sched) {
+  return launch_gemm256(a, b, 1 + (int)sched, stats);
 }

@@ -49,7 +49,7 @@ class _FusedBNAct(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, res, gamma, beta, eps, relu_mode, momentum,
-                r_mean, r_var, comm=None, unbias=None):
+                r_mean, r_var, comm=None, unbias=None, pre_stats=None):
         C = x.shape[1]
         count = x.numel() // C
         world = comm.size if comm is not None else 1
@@ -60,8 +60,15 @@ class _FusedBNAct(torch.autograd.Function):
             unbias = total / max(total - 1, 1)
         if x.is_cuda:
             ext = ops.require()
-            buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
-            ext.bn_stats_into(x, buf)
+            if pre_stats is not None:
+                # producer kernel (conv/GEMM epilogue) already reduced
+                # (sum y | sum y^2) into [16, 2C] partials — the
+                # standalone bn_reduce read pass over x is skipped
+                buf = pre_stats.sum(0)
+            else:
+                buf = torch.empty(2 * C, dtype=torch.float32,
+                                  device=x.device)
+                ext.bn_stats_into(x, buf)
             if world > 1:
                 import torch.distributed as dist
                 dist.all_reduce(buf, group=comm.group)
@@ -160,7 +167,7 @@ class _FusedBNAct(torch.autograd.Function):
             gres = gyf.to(x.dtype) if ctx.has_res else None
         g_res = gres if ctx.has_res else None
         return (gx, g_res, ggamma.to(gamma.dtype), gbeta.to(beta.dtype),
-                None, None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None)
 
 
 class FusedBatchNorm2d(nn.Module):
@@ -221,12 +228,12 @@ class FusedBatchNorm2d(nn.Module):
         m = self.momentum if self.momentum is not None else 0.1
         return self.running_mean, self.running_var, m
 
-    def _train_forward(self, x, res, relu_mode):
+    def _train_forward(self, x, res, relu_mode, pre_stats=None):
         gamma, beta = self._gamma_beta(x)
         r_mean, r_var, m = self._running()
         y = _FusedBNAct.apply(x, res, gamma, beta, self.eps, relu_mode, m,
                               r_mean, r_var, self.comm,
-                              self._running_unbias)
+                              self._running_unbias, pre_stats)
         if self.track_running_stats:
             self.num_batches_tracked += 1
         return y
@@ -248,15 +255,16 @@ class FusedBatchNorm2d(nn.Module):
             y = torch.relu(y)
         return y.to(x.dtype)
 
-    def forward(self, x):
+    def forward(self, x, pre_stats=None):
         if self.training or not self.track_running_stats:
-            return self._train_forward(x, None, self._relu_mode)
+            return self._train_forward(x, None, self._relu_mode,
+                                       pre_stats)
         return self._eval_forward(x, None, self._relu_mode)
 
-    def forward_add_relu(self, x, res):
+    def forward_add_relu(self, x, res, pre_stats=None):
         """y = relu(bn(x) + res) in one kernel pass (ResNet residual)."""
         if self.training or not self.track_running_stats:
-            return self._train_forward(x, res, 2)
+            return self._train_forward(x, res, 2, pre_stats)
         return self._eval_forward(x, res, 2)
 
     def extra_repr(self):

@@ -139,3 +139,74 @@ class FusedConv3x3(nn.Module):
     def extra_repr(self):
         return '{}, {}, kernel_size=3, stride={} (implicit GEMM)'.format(
             self.in_channels, self.out_channels, self.stride)
+
+
+class _Conv1x1StatsFn(torch.autograd.Function):
+    """_Conv1x1Fn variant that also returns the producer-fused BN stats
+    partials (empty tensor when the dispatched kernel can't fuse them)."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        N, C, H, W = x.shape
+        x2 = x.permute(0, 2, 3, 1).reshape(-1, C)
+        w2 = w.reshape(w.shape[0], C)
+        y2, stats = ops.gemm_bt_with_stats(x2, w2)
+        y = y2.view(N, H, W, -1).permute(0, 3, 1, 2)
+        ctx.save_for_backward(x, w)
+        if stats is None:
+            stats = x.new_empty(0, dtype=torch.float32)
+        ctx.mark_non_differentiable(stats)
+        return y, stats
+
+    @staticmethod
+    def backward(ctx, gy, g_stats):
+        gx, gw = _Conv1x1Fn.backward(ctx, gy)
+        return gx, gw
+
+
+class _Conv3x3StatsFn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        y, stats = ops.conv3x3_with_stats(x, w)
+        if stats is None:
+            stats = x.new_empty(0, dtype=torch.float32)
+        ctx.mark_non_differentiable(stats)
+        return y, stats
+
+    @staticmethod
+    def backward(ctx, gy, g_stats):
+        gx, gw = _Conv3x3Fn.backward(ctx, gy)
+        return gx, gw
+
+
+def _stats_or_none(st):
+    return st if st.numel() else None
+
+
+def _conv1x1_forward_stats(self, x):
+    """(y, BN-stats partials or None) — the fused conv->bn fast path."""
+    if x.is_cuda and x.dim() == 4 \
+            and x.is_contiguous(memory_format=torch.channels_last) \
+            and not x.is_contiguous():
+        y, st = _Conv1x1StatsFn.apply(x, self.weight)
+        return y, _stats_or_none(st)
+    return self.forward(x), None
+
+
+FusedConv1x1.forward_stats = _conv1x1_forward_stats
+
+
+def _conv3x3_forward_stats(self, x):
+    """(y, BN-stats partials or None) — the fused conv->bn fast path."""
+    if (self.stride == 1 and x.is_cuda and x.dim() == 4
+            and x.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and not x.is_contiguous()):
+        y, st = _Conv3x3StatsFn.apply(x, self.weight)
+        return y, _stats_or_none(st)
+    return self.forward(x), None
+
+
+FusedConv3x3.forward_stats = _conv3x3_forward_stats

@@ -45,12 +45,17 @@ class Bottleneck(nn.Module):
     def forward(self, x):
         identity = x
         if self.fused_ops:
-            # BN+ReLU and BN+Add+ReLU are single CDNA4 kernels
-            out = self.bn1(self.conv1(x))
-            out = self.bn2(self.conv2(out))
+            # BN+ReLU and BN+Add+ReLU are single CDNA4 kernels; the conv
+            # epilogues hand their (sum y | sum y^2) partials straight to
+            # the following BN, skipping its standalone stats read pass
+            y, st = self.conv1.forward_stats(x)
+            out = self.bn1(y, pre_stats=st)
+            y, st = self.conv2.forward_stats(out)
+            out = self.bn2(y, pre_stats=st)
             if self.downsample is not None:
                 identity = self.downsample(x)
-            return self.bn3.forward_add_relu(self.conv3(out), identity)
+            y, st = self.conv3.forward_stats(out)
+            return self.bn3.forward_add_relu(y, identity, pre_stats=st)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))

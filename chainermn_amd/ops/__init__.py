@@ -358,3 +358,51 @@ def conv3x3(x, w):
     if choice == 'mfma':
         return require().conv3x3s1_fwd_bf16(x, w)
     return F.conv2d(x, w, padding=1)
+
+
+def conv3x3_with_stats(x, w):
+    """Like conv3x3 but also returns the fused BN-stats partials
+    ([16, 2*Cout] fp32, sum over dim 0 = (sum y | sum y^2)) when the
+    dispatched kernel produces them, else (y, None)."""
+    import torch.nn.functional as F
+    eligible = (x.is_cuda and x.dtype == torch.bfloat16
+                and w.dtype == torch.bfloat16 and x.dim() == 4
+                and x.shape[1] % 64 == 0 and w.shape[0] % 64 == 0
+                and (x.shape[0] * x.shape[2] * x.shape[3]) % 256 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and w.is_contiguous(memory_format=torch.channels_last))
+    if not eligible:
+        return F.conv2d(x, w, padding=1), None
+    key = ('c3', tuple(x.shape), w.shape[0])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        ext = require()
+        choice = _decide_multi(key, {
+            'mfma': lambda: ext.conv3x3s1_fwd_bf16(x, w),
+            'miopen': lambda: F.conv2d(x, w, padding=1)})
+    if choice == 'mfma':
+        stats = torch.zeros(16, 2 * w.shape[0], dtype=torch.float32,
+                            device=x.device)
+        return require().conv3x3s1_fwd_bf16(x, w, stats), stats
+    return F.conv2d(x, w, padding=1), None
+
+
+def gemm_bt_with_stats(a, b):
+    """Like gemm_bt but also returns fused per-column BN-stats partials
+    when the 256-tile phased kernel is the dispatch winner (the 1x1-conv
+    forward path), else (d, None)."""
+    eligible = (a.is_cuda and a.dtype == torch.bfloat16
+                and a.shape[0] % 128 == 0 and b.shape[0] % 128 == 0
+                and a.shape[1] % 64 == 0 and a.is_contiguous()
+                and b.is_contiguous())
+    if not eligible:
+        return a @ b.t(), None
+    key = (a.shape[0], b.shape[0], a.shape[1])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        return gemm_bt(a, b), None  # decide on the plain path first
+    if choice == 'mfma256':
+        stats = torch.zeros(16, 2 * b.shape[0], dtype=torch.float32,
+                            device=a.device)
+        return require().gemm_bt256p_stats_bf16(a, b, stats, 3), stats
+    return gemm_bt(a, b), None

@@ -775,3 +775,66 @@ def test_fused_conv3x3_stride2_fallback():
     y = conv(x)
     yr = torch.nn.functional.conv2d(x, conv.weight, stride=2, padding=1)
     assert torch.equal(y.float(), yr.float())
+
+
+def test_producer_fused_bn_stats_match_bn_reduce():
+    """Conv/GEMM epilogue stats partials == the standalone bn_stats pass."""
+    torch.manual_seed(57)
+    # conv3x3 epilogue (all three wave geometries)
+    for N, C, H, W, Cout in [(4, 64, 8, 8, 64), (2, 128, 16, 16, 128),
+                             (1, 64, 16, 16, 256)]:
+        x = torch.randn(N, C, H, W, device=DEV, dtype=torch.bfloat16
+                        ).to(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, C, 3, 3, device=DEV, dtype=torch.bfloat16)
+             * 0.05).to(memory_format=torch.channels_last)
+        stats = torch.zeros(16, 2 * Cout, device=DEV)
+        y = _ext().conv3x3s1_fwd_bf16(x, w, stats)
+        ref = torch.empty(2 * Cout, device=DEV)
+        _ext().bn_stats_into(y, ref)
+        got = stats.sum(0)
+        assert torch.allclose(got, ref, rtol=1e-3, atol=2e-2), \
+            (N, C, H, W, Cout, (got - ref).abs().max())
+    # 256-tile phased GEMM epilogue
+    a = torch.randn(512, 128, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(256, 128, device=DEV, dtype=torch.bfloat16) * 0.05
+    stats = torch.zeros(16, 512, device=DEV)
+    d = _ext().gemm_bt256p_stats_bf16(a, b, stats, 3)
+    ref = torch.empty(512, device=DEV)
+    _ext().bn_stats_into(d, ref)
+    got = stats.sum(0)
+    assert torch.allclose(got, ref, rtol=1e-3, atol=2e-2), \
+        (got - ref).abs().max()
+
+
+def test_bn_pre_stats_path_matches_plain():
+    """FusedBatchNorm2d(pre_stats=...) == the standalone-stats path,
+    forward and backward."""
+    from chainermn_amd.links import FusedConv3x3
+    from chainermn_amd.links.fused_batch_norm import FusedBNReLU2d
+    torch.manual_seed(58)
+    conv = FusedConv3x3(64, 128).to(DEV, torch.bfloat16)
+    with torch.no_grad():
+        conv.weight.mul_(0.05)
+    conv = conv.to(memory_format=torch.channels_last)
+    bn = FusedBNReLU2d(128).to(DEV)
+    bn2 = FusedBNReLU2d(128).to(DEV)
+    bn2.load_state_dict(bn.state_dict())
+    x32 = torch.randn(4, 64, 16, 16, device=DEV)
+    x = x32.to(torch.bfloat16).to(memory_format=torch.channels_last
+                                  ).requires_grad_(True)
+    xr = x32.to(torch.bfloat16).to(memory_format=torch.channels_last
+                                   ).requires_grad_(True)
+    y, st = conv.forward_stats(x)
+    assert st is not None  # the custom kernel must have been dispatched
+    out = bn(y, pre_stats=st)
+    ref = bn2(conv(xr))
+    assert torch.allclose(out.float(), ref.float(), atol=3e-2), \
+        (out.float() - ref.float()).abs().max()
+    g = torch.randn_like(ref)
+    out.backward(g)
+    ref.backward(g)
+    assert torch.allclose(x.grad.float(), xr.grad.float(), atol=3e-2)
+    assert torch.allclose(bn.weight.grad, bn2.weight.grad,
+                          rtol=1e-2, atol=1e-2)
+    assert torch.allclose(bn.running_var, bn2.running_var,
+                          rtol=1e-3, atol=1e-4)
