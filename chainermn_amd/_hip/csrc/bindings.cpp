@@ -60,6 +60,7 @@ torch::Tensor gemm_bt256p_stats_bf16(torch::Tensor a, torch::Tensor b,
                                      torch::Tensor stats, long sched);
 torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
                                  torch::Tensor stats);
+torch::Tensor conv3x3_wgrad_bf16(torch::Tensor x, torch::Tensor gy);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -114,6 +115,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "implicit-GEMM 3x3/s1/p1 conv forward (phased MFMA); optional "
         "fused BN-stats partials [16, 2*Cout]",
         py::arg("x"), py::arg("w"), py::arg("stats") = py::none());
+  m.def("conv3x3_wgrad_bf16", &conv3x3_wgrad_bf16,
+        "implicit-GEMM 3x3/s1/p1 conv weight-grad (pair-transposed LDS "
+        "staging, split-M fp32 atomics)");
   m.def("gemm_bt256p_stats_bf16", &gemm_bt256p_stats_bf16,
         "256-tile phased GEMM with fused per-column BN-stats partials",
         py::arg("a"), py::arg("b"), py::arg("stats"),

@@ -105,9 +105,7 @@ class _Conv3x3Fn(torch.autograd.Function):
         wt = w.flip(2, 3).permute(1, 0, 2, 3).contiguous(
             memory_format=torch.channels_last)
         gx = ops.conv3x3(gy, wt)
-        _, gw, _ = torch.ops.aten.convolution_backward(
-            gy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-            [False, True, False])
+        gw = ops.conv3x3_wgrad(x, gy, w)
         return gx, gw
 
 

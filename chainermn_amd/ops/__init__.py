@@ -406,3 +406,31 @@ def gemm_bt_with_stats(a, b):
                             device=a.device)
         return require().gemm_bt256p_stats_bf16(a, b, stats, 3), stats
     return gemm_bt(a, b), None
+
+
+def conv3x3_wgrad(x, gy, w):
+    """3x3/s1/p1 conv weight gradient: per-shape autotune between the
+    hand-written split-M implicit-GEMM kernel (wgrad3x3.hip) and
+    MIOpen's wrw solvers via aten.convolution_backward."""
+    def miopen():
+        return torch.ops.aten.convolution_backward(
+            gy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [False, True, False])[1]
+    eligible = (x.is_cuda and x.dtype == torch.bfloat16
+                and gy.dtype == torch.bfloat16
+                and x.shape[1] % 64 == 0 and gy.shape[1] % 64 == 0
+                and x.shape[2] * x.shape[3] >= 2
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and gy.is_contiguous(memory_format=torch.channels_last))
+    if not eligible:
+        return miopen()
+    key = ('wg', tuple(x.shape), gy.shape[1])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        ext = require()
+        choice = _decide_multi(key, {
+            'mfma': lambda: ext.conv3x3_wgrad_bf16(x, gy),
+            'miopen': miopen})
+    if choice == 'mfma':
+        return require().conv3x3_wgrad_bf16(x, gy)
+    return miopen()

@@ -838,3 +838,28 @@ def test_bn_pre_stats_path_matches_plain():
                           rtol=1e-2, atol=1e-2)
     assert torch.allclose(bn.running_var, bn2.running_var,
                           rtol=1e-3, atol=1e-4)
+
+
+def test_conv3x3_wgrad_vs_reference():
+    """Hand-written wgrad kernel vs aten.convolution_backward (MIOpen),
+    both against an fp32 conv_backward reference."""
+    torch.manual_seed(61)
+    for N, C, H, W, Co in [(4, 64, 8, 8, 64), (2, 128, 16, 16, 128),
+                           (1, 64, 16, 16, 256), (3, 64, 7, 7, 64),
+                           (2, 192, 9, 9, 64)]:
+        x32 = torch.randn(N, C, H, W, device=DEV)
+        gy32 = torch.randn(N, Co, H, W, device=DEV) * 0.1
+        x = x32.to(torch.bfloat16).to(memory_format=torch.channels_last)
+        gy = gy32.to(torch.bfloat16).to(
+            memory_format=torch.channels_last)
+        gw = _ext().conv3x3_wgrad_bf16(x, gy)
+        # fp32 reference on the same bf16-rounded inputs
+        w_ref = torch.zeros(Co, C, 3, 3, device=DEV, requires_grad=True)
+        yr = torch.nn.functional.conv2d(x.float(), w_ref, padding=1)
+        yr.backward(gy.float())
+        ref = w_ref.grad
+        err = (gw.float() - ref).abs().max()
+        mag = ref.abs().max()
+        assert err <= 0.02 * mag + 0.05, (N, C, H, W, Co, err, mag)
+        assert gw.is_contiguous(memory_format=torch.channels_last)
+        assert gw.dtype == torch.bfloat16

@@ -36,3 +36,16 @@ for N, C, H, W, Cout in [(512, 64, 56, 56, 64), (512, 128, 28, 28, 128),
     d1 = t(lambda: ops.require().conv3x3s1_fwd_bf16(x, w))
     d2 = t(lambda: F.conv2d(x, w, padding=1))
     print(f"  N{N} C{C} {H}x{W} -> {Cout}: ours {fl/d1/1e12:6.1f} TF ({d1*1e6:6.1f}us) | miopen {fl/d2/1e12:6.1f} TF ({d2*1e6:6.1f}us)")
+
+print("\n3x3 conv WGRAD: ours vs MIOpen wrw (aten.convolution_backward):")
+for N, C, H, W, Co in [(512, 64, 56, 56, 64), (512, 128, 28, 28, 128),
+                       (512, 256, 14, 14, 256), (512, 512, 7, 7, 512)]:
+    x = (torch.rand(N, C, H, W, device='cuda', dtype=torch.bfloat16)*2-1).to(memory_format=torch.channels_last)
+    gy = ((torch.rand(N, Co, H, W, device='cuda', dtype=torch.bfloat16)*2-1)*0.1).to(memory_format=torch.channels_last)
+    w = torch.zeros(Co, C, 3, 3, device='cuda', dtype=torch.bfloat16).to(memory_format=torch.channels_last)
+    fl = 2.0*N*H*W*Co*9*C
+    d1 = t(lambda: ops.require().conv3x3_wgrad_bf16(x, gy))
+    d2 = t(lambda: torch.ops.aten.convolution_backward(
+        gy, x, w, None, [1,1], [1,1], [1,1], False, [0,0], 1,
+        [False, True, False])[1])
+    print(f"  N{N} C{C} {H}x{W} Co{Co}: ours {fl/d1/1e12:6.1f} TF ({d1*1e6:7.1f}us) | miopen {fl/d2/1e12:6.1f} TF ({d2*1e6:7.1f}us)")
