@@ -299,9 +299,11 @@ class FusedDataParallelSGD(object):
         # grad_pack: leave p.grad to autograd during backward and gather
         # with ONE batched kernel at step() instead of binding arena
         # views (which makes AccumulateGrad launch one add kernel per
-        # param per step — measured 3.4 ms/step on ResNet-50 b512).
-        # Default on for GPU non-bucketed modes (bucketed needs the
-        # views for in-place per-bucket all-reduce during backward).
+        # param per step).  Default on for GPU non-bucketed modes
+        # (bucketed needs the views for in-place per-bucket all-reduce
+        # during backward).  Pass grad_pack=False under hipGraph capture:
+        # the per-step pointer upload is a pageable H2D copy, which
+        # capture rejects.
         if grad_pack is None:
             grad_pack = (overlap != 'bucketed'
                          and self.state.device.type == 'cuda')
@@ -419,10 +421,11 @@ class FusedDataParallelAdam(FusedDataParallelSGD):
 
     def __init__(self, model, comm, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=0.0, adamw=False, overlap=True,
-                 grad_dtype=torch.bfloat16, bucket_mb=25):
+                 grad_dtype=torch.bfloat16, bucket_mb=25, grad_pack=None):
         super().__init__(model, comm, lr, momentum=0.0,
                          weight_decay=weight_decay, overlap=overlap,
-                         grad_dtype=grad_dtype, bucket_mb=bucket_mb)
+                         grad_dtype=grad_dtype, bucket_mb=bucket_mb,
+                         grad_pack=grad_pack)
         n = self.state.plan.total
         dev = self.state.device
         self.betas = betas
@@ -462,10 +465,11 @@ class FusedDataParallelLARS(FusedDataParallelSGD):
 
     def __init__(self, model, comm, lr, momentum=0.9, weight_decay=5e-5,
                  eta=0.001, eps=1e-9, overlap=True,
-                 grad_dtype=torch.bfloat16, bucket_mb=25):
+                 grad_dtype=torch.bfloat16, bucket_mb=25, grad_pack=None):
         super().__init__(model, comm, lr, momentum=momentum,
                          weight_decay=weight_decay, overlap=overlap,
-                         grad_dtype=grad_dtype, bucket_mb=bucket_mb)
+                         grad_dtype=grad_dtype, bucket_mb=bucket_mb,
+                         grad_pack=grad_pack)
         if self.v is None:
             n = self.state.plan.total
             self.v = torch.zeros(n, dtype=torch.float32,
