@@ -61,6 +61,8 @@ torch::Tensor gemm_bt256p_stats_bf16(torch::Tensor a, torch::Tensor b,
 torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
                                  torch::Tensor stats);
 torch::Tensor conv3x3_wgrad_bf16(torch::Tensor x, torch::Tensor gy);
+torch::Tensor gemm_btw_bf16(torch::Tensor a, torch::Tensor b,
+                            torch::Tensor stats);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -118,6 +120,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_wgrad_bf16", &conv3x3_wgrad_bf16,
         "implicit-GEMM 3x3/s1/p1 conv weight-grad (pair-transposed LDS "
         "staging, split-M fp32 atomics)");
+  m.def("gemm_btw_bf16",
+        [](torch::Tensor a, torch::Tensor b,
+           c10::optional<torch::Tensor> stats) {
+          return gemm_btw_bf16(
+              a, b, stats.has_value() ? *stats : torch::Tensor());
+        },
+        "narrow-N phased MFMA GEMM (BN 128/64): D = A @ B^T; optional "
+        "fused BN-stats partials",
+        py::arg("a"), py::arg("b"), py::arg("stats") = py::none());
   m.def("gemm_bt256p_stats_bf16", &gemm_bt256p_stats_bf16,
         "256-tile phased GEMM with fused per-column BN-stats partials",
         py::arg("a"), py::arg("b"), py::arg("stats"),

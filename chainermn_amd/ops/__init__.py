@@ -300,9 +300,13 @@ def gemm_bt(a, b):
                  'blas': lambda: a @ b.t()}
         if a.shape[0] % 256 == 0 and b.shape[0] % 256 == 0:
             cands['mfma256'] = lambda: ext.gemm_bt256p_bf16(a, b, 3)
+        elif a.shape[0] % 256 == 0 and b.shape[0] % 64 == 0:
+            cands['mfmaw'] = lambda: ext.gemm_btw_bf16(a, b)
         choice = _decide_multi(key, cands)
     if choice == 'mfma256':
         return require().gemm_bt256p_bf16(a, b, 3)
+    if choice == 'mfmaw':
+        return require().gemm_btw_bf16(a, b)
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b)
     return a @ b.t()
@@ -405,6 +409,10 @@ def gemm_bt_with_stats(a, b):
         stats = torch.zeros(16, 2 * b.shape[0], dtype=torch.float32,
                             device=a.device)
         return require().gemm_bt256p_stats_bf16(a, b, stats, 3), stats
+    if choice == 'mfmaw':
+        stats = torch.zeros(16, 2 * b.shape[0], dtype=torch.float32,
+                            device=a.device)
+        return require().gemm_btw_bf16(a, b, stats), stats
     return gemm_bt(a, b), None
 
 

@@ -863,3 +863,25 @@ def test_conv3x3_wgrad_vs_reference():
         assert err <= 0.02 * mag + 0.05, (N, C, H, W, Co, err, mag)
         assert gw.is_contiguous(memory_format=torch.channels_last)
         assert gw.dtype == torch.bfloat16
+
+
+def test_mfma_gemm_btw_vs_matmul():
+    """Narrow-N phased GEMM (BN 128/64) numerics + fused stats."""
+    torch.manual_seed(63)
+    for M, N, K in [(256, 128, 64), (512, 64, 128), (768, 192, 320),
+                    (256, 128, 1024), (512, 320, 64)]:
+        a = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+        b = torch.randn(N, K, device=DEV, dtype=torch.bfloat16)
+        a[0, :] += torch.arange(K, device=DEV, dtype=torch.bfloat16) * 0.01
+        b[1, :] -= torch.arange(K, device=DEV, dtype=torch.bfloat16) * 0.02
+        ref = a.float() @ b.float().t()
+        mag = ref.abs().max()
+        d = _ext().gemm_btw_bf16(a, b)
+        assert (d.float() - ref).abs().max() <= 0.02 * mag + 0.1, (M, N, K)
+        stats = torch.zeros(16, 2 * N, device=DEV)
+        d2 = _ext().gemm_btw_bf16(a, b, stats)
+        assert torch.equal(d, d2)
+        sref = torch.empty(2 * N, device=DEV)
+        _ext().bn_stats_into(d, sref)
+        assert torch.allclose(stats.sum(0), sref, rtol=1e-3, atol=2e-2), \
+            (M, N, K, (stats.sum(0) - sref).abs().max())

@@ -49,3 +49,17 @@ for N, C, H, W, Co in [(512, 64, 56, 56, 64), (512, 128, 28, 28, 128),
         gy, x, w, None, [1,1], [1,1], [1,1], False, [0,0], 1,
         [False, True, False])[1])
     print(f"  N{N} C{C} {H}x{W} Co{Co}: ours {fl/d1/1e12:6.1f} TF ({d1*1e6:7.1f}us) | miopen {fl/d2/1e12:6.1f} TF ({d2*1e6:7.1f}us)")
+
+print("\nnarrow-N phased GEMM (BN 128/64) vs 128-tile vs hipBLASLt:")
+for M, N, K in [(401408, 128, 512), (401408, 64, 256), (100352, 128, 512),
+                (1605632, 64, 576), (100352, 192, 1024)]:
+    a = (torch.rand(M, K, device='cuda', dtype=torch.bfloat16)*2-1)
+    b = (torch.rand(N, K, device='cuda', dtype=torch.bfloat16)*2-1)
+    fl = 2.0*M*N*K
+    dw = t(lambda: ops.require().gemm_btw_bf16(a, b))
+    db_ = t(lambda: a @ b.t())
+    line = f"  M{M} N{N} K{K}: wide {fl/dw/1e12:6.1f} TF | blas {fl/db_/1e12:6.1f} TF"
+    if N % 128 == 0:
+        d1 = t(lambda: ops.require().gemm_bt_bf16(a, b))
+        line += f" | 128t {fl/d1/1e12:6.1f} TF"
+    print(line)
