@@ -235,27 +235,9 @@ def _time_gpu(fn, n=5):
 
 def _decide_multi(key, candidates):
     """Pick the fastest of named candidate callables for one shape,
-    identically on every rank (see _decide)."""
-    import torch.distributed as dist
-    synced = (dist.is_available() and dist.is_initialized()
-              and dist.get_world_size() > 1)
-    if synced and dist.get_rank() != 0:
-        lst = [None]
-        dist.broadcast_object_list(lst, src=0)
-        choice = lst[0]
-    else:
-        times = {name: _time_gpu(fn) for name, fn in candidates.items()}
-        choice = min(times, key=times.get)
-        if synced:
-            dist.broadcast_object_list([choice], src=0)
-    _GEMM_CHOICE[key] = choice
-    return choice
+    identically on every rank.
 
-
-def _decide(key, mfma_fn, blas_fn):
-    """Pick mfma-vs-blas for one shape, identically on every rank.
-
-    Rank 0 times both candidates (this calls torch.cuda.synchronize, so
+    Rank 0 times the candidates (this calls torch.cuda.synchronize, so
     the one-time cost per new shape stalls any overlapped side-stream
     work once); the other ranks receive the choice over the CPU object
     channel instead of timing locally — local timing jitter would let
@@ -271,9 +253,8 @@ def _decide(key, mfma_fn, blas_fn):
         dist.broadcast_object_list(lst, src=0)
         choice = lst[0]
     else:
-        t_ours = _time_gpu(mfma_fn)
-        t_blas = _time_gpu(blas_fn)
-        choice = 'mfma' if t_ours < t_blas else 'blas'
+        times = {name: _time_gpu(fn) for name, fn in candidates.items()}
+        choice = min(times, key=times.get)
         if synced:
             dist.broadcast_object_list([choice], src=0)
     _GEMM_CHOICE[key] = choice
