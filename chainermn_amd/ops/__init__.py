@@ -311,9 +311,14 @@ def gemm_nn(a, b):
         if a.shape[0] % 256 == 0 and b.shape[1] % 256 == 0:
             cands['mfma256'] = lambda: ext.gemm_bt256p_bf16(
                 a, b.t().contiguous(), 3)
+        elif a.shape[0] % 256 == 0 and b.shape[1] % 64 == 0:
+            cands['mfmaw'] = lambda: ext.gemm_btw_bf16(
+                a, b.t().contiguous())
         choice = _decide_multi(key, cands)
     if choice == 'mfma256':
         return require().gemm_bt256p_bf16(a, b.t().contiguous(), 3)
+    if choice == 'mfmaw':
+        return require().gemm_btw_bf16(a, b.t().contiguous())
     if choice == 'mfma':
         return require().gemm_bt_bf16(a, b.t().contiguous())
     return a @ b

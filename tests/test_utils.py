@@ -56,3 +56,29 @@ def test_gemm_wrappers_cpu_fallback():
     assert torch.allclose(ops.gemm_bt(a, b), a @ b.t(), atol=1e-5)
     c = torch.randn(16, 24)
     assert torch.allclose(ops.gemm_nn(a, c), a @ c, atol=1e-5)
+
+
+def test_conv_ops_cpu_fallbacks():
+    """ops.conv3x3 / conv3x3_with_stats / conv3x3_wgrad fall back to
+    torch ops for CPU tensors (GPU kernels dispatch only when eligible)."""
+    import torch
+    import torch.nn.functional as F
+    from chainermn_amd import ops
+    torch.manual_seed(5)
+    x = torch.randn(2, 8, 6, 6)
+    w = torch.randn(8, 8, 3, 3) * 0.1
+    y = ops.conv3x3(x, w)
+    assert torch.allclose(y, F.conv2d(x, w, padding=1))
+    y2, st = ops.conv3x3_with_stats(x, w)
+    assert st is None and torch.equal(y, y2)
+    gy = torch.randn_like(y)
+    gw = ops.conv3x3_wgrad(x, gy, w)
+    ref = torch.ops.aten.convolution_backward(
+        gy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+        [False, True, False])[1]
+    assert torch.allclose(gw, ref)
+    # gemm stats fallback: CPU / undecided shapes return stats=None
+    a = torch.randn(128, 64)
+    b = torch.randn(128, 64)
+    d, st = ops.gemm_bt_with_stats(a, b)
+    assert st is None and torch.allclose(d, a @ b.t(), atol=1e-5)
