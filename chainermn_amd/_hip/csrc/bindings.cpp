@@ -63,6 +63,10 @@ torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
 torch::Tensor conv3x3_wgrad_bf16(torch::Tensor x, torch::Tensor gy);
 torch::Tensor gemm_btw_bf16(torch::Tensor a, torch::Tensor b,
                             torch::Tensor stats);
+torch::Tensor gemm_btw_acc_bf16(torch::Tensor a, torch::Tensor b,
+                                torch::Tensor out);
+torch::Tensor gemm_bt256p_acc_bf16(torch::Tensor a, torch::Tensor b,
+                                   torch::Tensor out);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -129,6 +133,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "narrow-N phased MFMA GEMM (BN 128/64): D = A @ B^T; optional "
         "fused BN-stats partials",
         py::arg("a"), py::arg("b"), py::arg("stats") = py::none());
+  m.def("gemm_btw_acc_bf16", &gemm_btw_acc_bf16,
+        "narrow-N phased GEMM, accumulate mode: out += A @ B^T");
+  m.def("gemm_bt256p_acc_bf16", &gemm_bt256p_acc_bf16,
+        "256-tile phased GEMM, accumulate mode: out += A @ B^T");
   m.def("gemm_bt256p_stats_bf16", &gemm_bt256p_stats_bf16,
         "256-tile phased GEMM with fused per-column BN-stats partials",
         py::arg("a"), py::arg("b"), py::arg("stats"),

@@ -167,7 +167,9 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256_bf16_kernel(
 // cluster (1 barrier per phase): consecutive phases of the two
 // co-resident waves can then pair a load segment against the partner's
 // MFMA segment instead of matrix||matrix.
-template <int SCHED>
+// ACC: accumulate into D (D += A@B^T) instead of overwriting — used to
+// fold the residual-junction gradient add into the dgrad epilogue.
+template <int SCHED, bool ACC = false>
 __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ D, float* __restrict__ SP, int M, int N, int K) {
@@ -359,8 +361,10 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
       const long row = row_base + m * 16 + rg;
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
-        const float v = acc[m][n][rg];
-        D[row * N + col0 + n * 16] = (short)c10::BFloat16(v).x;
+        float v = acc[m][n][rg];
+        short* dp2 = &D[row * N + col0 + n * 16];
+        if constexpr (ACC) v += float(*reinterpret_cast<c10::BFloat16*>(dp2));
+        *dp2 = (short)c10::BFloat16(v).x;
       }
     }
   }
@@ -439,4 +443,26 @@ torch::Tensor gemm_bt256p_bf16(torch::Tensor a, torch::Tensor b,
 torch::Tensor gemm_bt256p_stats_bf16(torch::Tensor a, torch::Tensor b,
                                      torch::Tensor stats, long sched) {
   return launch_gemm256(a, b, 1 + (int)sched, stats);
+}
+
+// out += a @ b^T in one pass (the residual-junction fold).  out must be
+// a contiguous bf16 [M, N] view of pre-existing gradient memory.
+torch::Tensor gemm_bt256p_acc_bf16(torch::Tensor a, torch::Tensor b,
+                                   torch::Tensor out) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16
+              && b.scalar_type() == at::kBFloat16
+              && out.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous()
+              && out.is_contiguous());
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(M % G2_BM == 0 && N % G2_BN == 0 && K % G2_BK == 0);
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int grid = (M / G2_BM) * (N / G2_BN);
+  hipLaunchKernelGGL((gemm_bt256p_bf16_kernel<3, true>), dim3(grid),
+                     dim3(G2_THREADS), 0, stream.stream(),
+                     (const short*)a.data_ptr(), (const short*)b.data_ptr(),
+                     (short*)out.data_ptr(), nullptr, M, N, K);
+  return out;
 }

@@ -45,7 +45,7 @@ __device__ __forceinline__ void gw_waitbar() {
     asm volatile("s_barrier" ::: "memory");
 }
 
-template <int WM, int WAVES_M, int WAVES_N>
+template <int WM, int WAVES_M, int WAVES_N, bool ACC = false>
 __global__ __launch_bounds__(GW_THREADS, 1) void gemm_btw_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
     short* __restrict__ D, float* __restrict__ SP, int M, int N, int K) {
@@ -224,9 +224,12 @@ __global__ __launch_bounds__(GW_THREADS, 1) void gemm_btw_kernel(
     for (int rg = 0; rg < 4; ++rg) {
       const long row = row_base + m * 16 + rg;
 #pragma unroll
-      for (int n = 0; n < 4; ++n)
-        D[row * N + col0 + n * 16] =
-            (short)c10::BFloat16(acc[m][n][rg]).x;
+      for (int n = 0; n < 4; ++n) {
+        float v = acc[m][n][rg];
+        short* dp2 = &D[row * N + col0 + n * 16];
+        if constexpr (ACC) v += float(*reinterpret_cast<c10::BFloat16*>(dp2));
+        *dp2 = (short)c10::BFloat16(v).x;
+      }
     }
   }
   if (SP != nullptr) {
@@ -273,4 +276,33 @@ torch::Tensor gemm_btw_bf16(torch::Tensor a, torch::Tensor b,
                        sp, M, N, K);
   }
   return d;
+}
+
+torch::Tensor gemm_btw_acc_bf16(torch::Tensor a, torch::Tensor b,
+                                torch::Tensor out) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16
+              && b.scalar_type() == at::kBFloat16
+              && out.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous()
+              && out.is_contiguous());
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(M % 256 == 0 && N % 64 == 0 && K % 64 == 0);
+  auto stream = c10::hip::getCurrentHIPStream();
+  const short* ap = (const short*)a.data_ptr();
+  const short* bp = (const short*)b.data_ptr();
+  short* dp = (short*)out.data_ptr();
+  if (N % 128 == 0) {
+    const long grid = ((long)M / 256) * (N / 128);
+    hipLaunchKernelGGL((gemm_btw_kernel<64, 4, 2, true>),
+                       dim3((unsigned)grid), dim3(GW_THREADS), 0,
+                       stream.stream(), ap, bp, dp, nullptr, M, N, K);
+  } else {
+    const long grid = ((long)M / 256) * (N / 64);
+    hipLaunchKernelGGL((gemm_btw_kernel<32, 8, 1, true>),
+                       dim3((unsigned)grid), dim3(GW_THREADS), 0,
+                       stream.stream(), ap, bp, dp, nullptr, M, N, K);
+  }
+  return out;
 }

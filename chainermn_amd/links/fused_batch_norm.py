@@ -49,7 +49,8 @@ class _FusedBNAct(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, res, gamma, beta, eps, relu_mode, momentum,
-                r_mean, r_var, comm=None, unbias=None, pre_stats=None):
+                r_mean, r_var, comm=None, unbias=None, pre_stats=None,
+                junction=None):
         C = x.shape[1]
         count = x.numel() // C
         world = comm.size if comm is not None else 1
@@ -115,6 +116,7 @@ class _FusedBNAct(torch.autograd.Function):
         ctx.count = total
         ctx.comm = comm if world > 1 else None
         ctx.has_res = res is not None
+        ctx.junction = junction
         return y
 
     @staticmethod
@@ -166,8 +168,13 @@ class _FusedBNAct(torch.autograd.Function):
                             + _shape(gbeta, x)) / count)).to(x.dtype)
             gres = gyf.to(x.dtype) if ctx.has_res else None
         g_res = gres if ctx.has_res else None
+        if ctx.junction is not None and g_res is not None:
+            # residual-junction handoff: the SAME tensor autograd routes
+            # to _ResidualEntry's identity slot; the block's conv1 dgrad
+            # may accumulate the main-path gradient into it in-kernel
+            ctx.junction['gres'] = g_res
         return (gx, g_res, ggamma.to(gamma.dtype), gbeta.to(beta.dtype),
-                None, None, None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None, None)
 
 
 class FusedBatchNorm2d(nn.Module):
@@ -228,12 +235,13 @@ class FusedBatchNorm2d(nn.Module):
         m = self.momentum if self.momentum is not None else 0.1
         return self.running_mean, self.running_var, m
 
-    def _train_forward(self, x, res, relu_mode, pre_stats=None):
+    def _train_forward(self, x, res, relu_mode, pre_stats=None,
+                       junction=None):
         gamma, beta = self._gamma_beta(x)
         r_mean, r_var, m = self._running()
         y = _FusedBNAct.apply(x, res, gamma, beta, self.eps, relu_mode, m,
                               r_mean, r_var, self.comm,
-                              self._running_unbias, pre_stats)
+                              self._running_unbias, pre_stats, junction)
         if self.track_running_stats:
             self.num_batches_tracked += 1
         return y
@@ -261,10 +269,10 @@ class FusedBatchNorm2d(nn.Module):
                                        pre_stats)
         return self._eval_forward(x, None, self._relu_mode)
 
-    def forward_add_relu(self, x, res, pre_stats=None):
+    def forward_add_relu(self, x, res, pre_stats=None, junction=None):
         """y = relu(bn(x) + res) in one kernel pass (ResNet residual)."""
         if self.training or not self.track_running_stats:
-            return self._train_forward(x, res, 2, pre_stats)
+            return self._train_forward(x, res, 2, pre_stats, junction)
         return self._eval_forward(x, res, 2)
 
     def extra_repr(self):

@@ -141,16 +141,23 @@ class FusedConv3x3(nn.Module):
 
 class _Conv1x1StatsFn(torch.autograd.Function):
     """_Conv1x1Fn variant that also returns the producer-fused BN stats
-    partials (empty tensor when the dispatched kernel can't fuse them)."""
+    partials (empty tensor when the dispatched kernel can't fuse them).
+
+    ``junction`` (a dict or None) is the residual-junction handoff: when
+    the SAME block's bn3 backward has deposited the identity-path
+    gradient there, this conv's dgrad GEMM accumulates into that tensor
+    in its epilogue (ops.gemm_nn_accum) and marks the junction 'fused' —
+    _ResidualEntry.backward then skips the engine's separate add."""
 
     @staticmethod
-    def forward(ctx, x, w):
+    def forward(ctx, x, w, junction=None):
         N, C, H, W = x.shape
         x2 = x.permute(0, 2, 3, 1).reshape(-1, C)
         w2 = w.reshape(w.shape[0], C)
         y2, stats = ops.gemm_bt_with_stats(x2, w2)
         y = y2.view(N, H, W, -1).permute(0, 3, 1, 2)
         ctx.save_for_backward(x, w)
+        ctx.junction = junction
         if stats is None:
             stats = x.new_empty(0, dtype=torch.float32)
         ctx.mark_non_differentiable(stats)
@@ -158,8 +165,53 @@ class _Conv1x1StatsFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy, g_stats):
+        j = ctx.junction
+        gres = j.pop('gres', None) if j is not None else None
+        if gres is not None:
+            x, w = ctx.saved_tensors
+            N, C, H, W = x.shape
+            Cout = w.shape[0]
+            gy2 = gy.contiguous(memory_format=torch.channels_last)                 .permute(0, 2, 3, 1).reshape(-1, Cout)
+            w2 = w.reshape(Cout, C)
+            gres2 = gres.permute(0, 2, 3, 1).reshape(-1, C)
+            if gres2.is_contiguous():
+                res2, folded = ops.gemm_nn_accum(gy2, w2, gres2)
+            else:
+                res2, folded = ops.gemm_nn(gy2, w2), False
+            if folded:
+                j['fused'] = True
+                gx = gres
+            else:
+                gx = res2.view(N, H, W, C).permute(0, 3, 1, 2)
+            _, gw, _ = torch.ops.aten.convolution_backward(
+                gy.contiguous(memory_format=torch.channels_last), x, w,
+                None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, False])
+            return gx, gw, None
         gx, gw = _Conv1x1Fn.backward(ctx, gy)
-        return gx, gw
+        return gx, gw, None
+
+
+class _ResidualEntry(torch.autograd.Function):
+    """Explicit fork of a residual block's input: backward adds the two
+    path gradients itself UNLESS the conv1-dgrad already folded the
+    main-path gradient into the identity gradient (junction['fused'])."""
+
+    @staticmethod
+    def forward(ctx, x, junction):
+        ctx.junction = junction
+        return x, x
+
+    @staticmethod
+    def backward(ctx, g_main, g_id):
+        j = ctx.junction
+        if j is not None and j.pop('fused', False):
+            return g_main, None  # g_id was folded into g_main in-kernel
+        if g_id is None:
+            return g_main, None
+        if g_main is None:
+            return g_id, None
+        return g_main + g_id, None
 
 
 class _Conv3x3StatsFn(torch.autograd.Function):
@@ -183,12 +235,12 @@ def _stats_or_none(st):
     return st if st.numel() else None
 
 
-def _conv1x1_forward_stats(self, x):
+def _conv1x1_forward_stats(self, x, junction=None):
     """(y, BN-stats partials or None) — the fused conv->bn fast path."""
     if x.is_cuda and x.dim() == 4 \
             and x.is_contiguous(memory_format=torch.channels_last) \
             and not x.is_contiguous():
-        y, st = _Conv1x1StatsFn.apply(x, self.weight)
+        y, st = _Conv1x1StatsFn.apply(x, self.weight, junction)
         return y, _stats_or_none(st)
     return self.forward(x), None
 

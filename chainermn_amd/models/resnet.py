@@ -15,6 +15,9 @@ import torch.nn as nn
 
 class Bottleneck(nn.Module):
     expansion = 4
+    # class-level switch for the residual-junction gradient fold
+    # (tests compare fold-on vs fold-off on identical kernel stacks)
+    _use_junction_fold = True
 
     def __init__(self, in_ch, width, stride=1, downsample=None,
                  norm_layer=nn.BatchNorm2d, fused_ops=False, comm=None):
@@ -47,15 +50,27 @@ class Bottleneck(nn.Module):
         if self.fused_ops:
             # BN+ReLU and BN+Add+ReLU are single CDNA4 kernels; the conv
             # epilogues hand their (sum y | sum y^2) partials straight to
-            # the following BN, skipping its standalone stats read pass
-            y, st = self.conv1.forward_stats(x)
+            # the following BN, skipping its standalone stats read pass.
+            # Identity blocks also fork the input through _ResidualEntry:
+            # in backward, conv1's dgrad GEMM accumulates the main-path
+            # gradient into the identity gradient in its epilogue, so
+            # the junction's separate grad-add never runs.
+            junction = None
+            x_main = x
+            if self.downsample is None and x.is_cuda \
+                    and self._use_junction_fold:
+                from chainermn_amd.links.fused_conv import _ResidualEntry
+                junction = {}
+                x_main, identity = _ResidualEntry.apply(x, junction)
+            y, st = self.conv1.forward_stats(x_main, junction=junction)
             out = self.bn1(y, pre_stats=st)
             y, st = self.conv2.forward_stats(out)
             out = self.bn2(y, pre_stats=st)
             if self.downsample is not None:
                 identity = self.downsample(x)
             y, st = self.conv3.forward_stats(out)
-            return self.bn3.forward_add_relu(y, identity, pre_stats=st)
+            return self.bn3.forward_add_relu(y, identity, pre_stats=st,
+                                             junction=junction)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))

@@ -428,3 +428,35 @@ def conv3x3_wgrad(x, gy, w):
     if choice == 'mfma':
         return require().conv3x3_wgrad_bf16(x, gy)
     return miopen()
+
+
+def gemm_nn_accum(a, b, out):
+    """``out += a @ b`` in one kernel pass when the phased MFMA kernels
+    are the dispatch winners for this shape (the residual-junction fold:
+    the dgrad GEMM writes its result accumulating into the identity
+    path's gradient, so the autograd engine's separate 3-pass add never
+    runs).  Returns (result, folded): folded=False means the caller owns
+    the add (first-call shape decision, blas/128-tile winners, CPU)."""
+    eligible = (a.is_cuda and a.dtype == torch.bfloat16
+                and out.dtype == torch.bfloat16
+                and a.shape[0] % 256 == 0 and b.shape[1] % 64 == 0
+                and a.shape[1] % 64 == 0 and a.is_contiguous()
+                and out.is_contiguous()
+                and out.shape[0] == a.shape[0]
+                and out.shape[1] == b.shape[1])
+    if not eligible:
+        return gemm_nn(a, b), False
+    key = ('nn', a.shape[0], b.shape[1], a.shape[1])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        return gemm_nn(a, b), False  # decide on the plain path first
+    if choice == 'mfma256' and b.shape[1] % 256 == 0:
+        return require().gemm_bt256p_acc_bf16(
+            a, b.t().contiguous(), out), True
+    if choice == 'mfmaw':
+        return require().gemm_btw_acc_bf16(
+            a, b.t().contiguous(), out), True
+    # hipBLASLt / 128-tile winners: accumulate through hipBLASLt's
+    # native beta=1 epilogue — still one pass, no separate add kernel
+    out.addmm_(a, b)
+    return out, True

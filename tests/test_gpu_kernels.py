@@ -885,3 +885,59 @@ def test_mfma_gemm_btw_vs_matmul():
         _ext().bn_stats_into(d, sref)
         assert torch.allclose(stats.sum(0), sref, rtol=1e-3, atol=2e-2), \
             (M, N, K, (stats.sum(0) - sref).abs().max())
+
+
+def test_residual_junction_fold_matches_unfused():
+    """The _ResidualEntry + conv1-dgrad accumulate fold must give the
+    same grads as the SAME fused stack with the fold disabled (second
+    step, when the per-shape GEMM choices are decided and the fold
+    engages).  Identical kernel stacks, so tolerances are tight."""
+    from chainermn_amd.models.resnet import Bottleneck
+    torch.manual_seed(77)
+    fused = torch.nn.Sequential(
+        Bottleneck(256, 64, fused_ops=True),
+        Bottleneck(256, 64, fused_ops=True),
+    ).to(DEV, torch.bfloat16).to(memory_format=torch.channels_last)
+    ref = torch.nn.Sequential(
+        Bottleneck(256, 64, fused_ops=True),
+        Bottleneck(256, 64, fused_ops=True),
+    ).to(DEV, torch.bfloat16).to(memory_format=torch.channels_last)
+    ref.load_state_dict(fused.state_dict())
+    try:
+        for step in range(2):
+            torch.manual_seed(300 + step)
+            x32 = torch.randn(4, 256, 16, 16, device=DEV) * 0.5
+            x = x32.to(torch.bfloat16).to(
+                memory_format=torch.channels_last).requires_grad_(True)
+            xr = x32.to(torch.bfloat16).to(
+                memory_format=torch.channels_last).requires_grad_(True)
+            Bottleneck._use_junction_fold = True
+            y = fused(x)
+            Bottleneck._use_junction_fold = False
+            yr = ref(xr)
+            # forward differs only by the atomic-order rounding of the
+            # producer-fused BN stats (+-1 ulp on mean/var)
+            fmag = yr.float().abs().max()
+            assert (y.float() - yr.float()).abs().max() \
+                <= 0.01 * fmag + 0.01, step
+            g = torch.randn_like(yr) * 0.1
+            Bottleneck._use_junction_fold = True
+            y.backward(g)
+            Bottleneck._use_junction_fold = False
+            yr.backward(g)
+            # the fold changes only the junction-add rounding order
+            # (fp32 accumulate + one bf16 round vs bf16 + bf16 add)
+            gmag = xr.grad.float().abs().max()
+            assert (x.grad.float() - xr.grad.float()).abs().max() \
+                <= 0.02 * gmag + 0.02, (step, 'input grad')
+            for (n1, p1), (n2, p2) in zip(fused.named_parameters(),
+                                          ref.named_parameters()):
+                if p1.grad is None or p2.grad is None:
+                    continue
+                pm = p2.grad.float().abs().max()
+                assert (p1.grad.float() - p2.grad.float()).abs().max() \
+                    <= 0.02 * pm + 0.02, (step, n1)
+            fused.zero_grad()
+            ref.zero_grad()
+    finally:
+        Bottleneck._use_junction_fold = True
