@@ -233,9 +233,12 @@ def _time_gpu(fn, n=5):
     return (time.perf_counter() - t0) / n
 
 
-def _decide_multi(key, candidates):
+def _decide_multi(key, candidates, margin=None):
     """Pick the fastest of named candidate callables for one shape,
-    identically on every rank.
+    identically on every rank.  ``margin`` maps a candidate name to a
+    factor its time is multiplied by before comparison (>1 = that
+    candidate must win by that factor — guards near-ties against
+    timing-noise flips to the slower kernel).
 
     Rank 0 times the candidates (this calls torch.cuda.synchronize, so
     the one-time cost per new shape stalls any overlapped side-stream
@@ -254,6 +257,10 @@ def _decide_multi(key, candidates):
         choice = lst[0]
     else:
         times = {name: _time_gpu(fn) for name, fn in candidates.items()}
+        if margin:
+            for name, f in margin.items():
+                if name in times:
+                    times[name] *= f
         choice = min(times, key=times.get)
         if synced:
             dist.broadcast_object_list([choice], src=0)
@@ -424,7 +431,7 @@ def conv3x3_wgrad(x, gy, w):
         ext = require()
         choice = _decide_multi(key, {
             'mfma': lambda: ext.conv3x3_wgrad_bf16(x, gy),
-            'miopen': miopen})
+            'miopen': miopen}, margin={'mfma': 1.05})
     if choice == 'mfma':
         return require().conv3x3_wgrad_bf16(x, gy)
     return miopen()
