@@ -192,6 +192,7 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
 
   bf16x8 af[2][MF];
   bf16x8 bfr[2][2];
+  bf16x8 bfr1[2][2];  // second B half (B_BOTH 2-phase loop only)
 
 #define C3_READ_A(buf, qa)                                                   \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
@@ -219,6 +220,24 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
   __builtin_amdgcn_s_setprio(0);
 #define C3_BAR() asm volatile("s_barrier" ::: "memory");
 
+#define C3_READ_B1(buf)                                                     \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+      _Pragma("unroll") for (int ni = 0; ni < 2; ++ni) {                     \
+    const int row = wc * 64 + 32 + ni * 16 + fr;                             \
+    bfr1[ks][ni] = *(const bf16x8*)(ldsB(buf) + row * 128                    \
+                                    + c3_swz(row, ks * 64 + kq * 16));       \
+  }
+#define C3_MFMA1(qa)                                                         \
+  __builtin_amdgcn_s_setprio(1);                                            \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                           \
+      _Pragma("unroll") for (int mi = 0; mi < MF; ++mi)                      \
+          _Pragma("unroll") for (int ni = 0; ni < 2; ++ni)                   \
+    acc[(qa) * MF + mi][2 + ni] =                                            \
+        __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             \
+            af[ks][mi], bfr1[ks][ni], acc[(qa) * MF + mi][2 + ni],           \
+            0, 0, 0);                                                        \
+  __builtin_amdgcn_s_setprio(0);
+
   const int nkt = 9 * (Cin / 64);
   // tap/channel counters for the tile being staged (one ahead); the
   // current tile's taps are never needed (its data is already in LDS)
@@ -231,59 +250,106 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
     }
   };
 
-  // prologue: all halves of tile 0 in stage order (A0, B0/B01, A1, B1);
-  // A0 AND B0 must land (for B_BOTH the B stage is 3rd-from-newest, so
-  // the allowance is SA, not SA+SB)
-  stage_a(0, 0, 0, 0, 0);
-  stage_b(0, 0, 0);
-  stage_a(0, 1, 0, 0, 0);
-  if constexpr (!B_BOTH) stage_b(0, 1, 0);
-  c3_waitbar<B_BOTH ? SA : SA + SB>();
-
-  for (int kt = 0; kt < nkt - 1; ++kt) {
-    const int buf = kt & 1;
-    const int nxt = buf ^ 1;
-    advance(sky, skx, sci0);
-    sk0 += 64;
-    // phase 0: quadrant (0,0); stage A-half0(kt+1)
-    stage_a(nxt, 0, sky, skx, sci0);
-    C3_READ_A(buf, 0)
-    C3_READ_B(buf, 0)
-    c3_waitbar<W_P0>();
-    C3_MFMA(0, 0)
-    // phase 1: quadrant (1,0); stage B-half0 (or both halves)
-    stage_b(nxt, 0, sk0);
-    C3_READ_A(buf, 1)
-    c3_waitbar<W_P1>();
-    C3_MFMA(1, 0)
-    // phase 2: quadrant (1,1); stage A-half1
-    stage_a(nxt, 1, sky, skx, sci0);
-    C3_READ_B(buf, 1)
-    C3_BAR()
-    C3_MFMA(1, 1)
-    // phase 3: quadrant (0,1); stage B-half1 (unless staged at p1)
-    if constexpr (!B_BOTH) stage_b(nxt, 1, sk0);
-    C3_READ_A(buf, 0)
-    c3_waitbar<W_P3>();
-    C3_MFMA(0, 1)
+  if constexpr (B_BOTH) {
+    // ---- 2-phase variant (Cout == 64): per phase 8 MFMAs instead of
+    // 4, half the barriers — the 4-phase schedule left this geometry
+    // wait-bound (8.2% MfmaUtil, profiles/conv3x3_pmc.txt).  Both B
+    // halves stay live in registers; both waits are vmcnt(2).
+    stage_a(0, 0, 0, 0, 0);
+    stage_b(0, 0, 0);
+    stage_a(0, 1, 0, 0, 0);
+    c3_waitbar<2>();  // A0 + B01 landed; A1 in flight
+    for (int kt = 0; kt < nkt - 1; ++kt) {
+      const int buf = kt & 1;
+      const int nxt = buf ^ 1;
+      advance(sky, skx, sci0);
+      sk0 += 64;
+      // phase A: quadrants (0,0)+(0,1); stage A-half0(kt+1)
+      stage_a(nxt, 0, sky, skx, sci0);
+      C3_READ_A(buf, 0)
+      C3_READ_B(buf, 0)
+      C3_READ_B1(buf)
+      c3_waitbar<2>();  // guard A1(kt); leaves A0(kt+1) in flight
+      C3_MFMA(0, 0)
+      C3_MFMA1(0)
+      // phase B: quadrants (1,0)+(1,1); stage B then A-half1 of kt+1
+      // (B BEFORE A1: the end-of-phase vmcnt(2) must leave only the A1
+      // glds outstanding — B01 is consumed first next tile)
+      stage_b(nxt, 0, sk0);
+      stage_a(nxt, 1, sky, skx, sci0);
+      C3_READ_A(buf, 1)
+      c3_waitbar<2>();  // guard A0/B01(kt+1); leaves A1(kt+1) in flight
+      C3_MFMA(1, 0)
+      C3_MFMA1(1)
+    }
+    {  // peeled last tile
+      const int buf = (nkt - 1) & 1;
+      C3_READ_A(buf, 0)
+      C3_READ_B(buf, 0)
+      C3_READ_B1(buf)
+      c3_waitbar<0>();  // guard A1(last)
+      C3_MFMA(0, 0)
+      C3_MFMA1(0)
+      C3_READ_A(buf, 1)
+      C3_BAR()
+      C3_MFMA(1, 0)
+      C3_MFMA1(1)
+    }
+  } else {
+  // prologue: all halves of tile 0 in stage order (A0, B0, A1, B1)
+    stage_a(0, 0, 0, 0, 0);
+    stage_b(0, 0, 0);
+    stage_a(0, 1, 0, 0, 0);
+    if constexpr (!B_BOTH) stage_b(0, 1, 0);
+    c3_waitbar<B_BOTH ? SA : SA + SB>();
+  
+    for (int kt = 0; kt < nkt - 1; ++kt) {
+      const int buf = kt & 1;
+      const int nxt = buf ^ 1;
+      advance(sky, skx, sci0);
+      sk0 += 64;
+      // phase 0: quadrant (0,0); stage A-half0(kt+1)
+      stage_a(nxt, 0, sky, skx, sci0);
+      C3_READ_A(buf, 0)
+      C3_READ_B(buf, 0)
+      c3_waitbar<W_P0>();
+      C3_MFMA(0, 0)
+      // phase 1: quadrant (1,0); stage B-half0 (or both halves)
+      stage_b(nxt, 0, sk0);
+      C3_READ_A(buf, 1)
+      c3_waitbar<W_P1>();
+      C3_MFMA(1, 0)
+      // phase 2: quadrant (1,1); stage A-half1
+      stage_a(nxt, 1, sky, skx, sci0);
+      C3_READ_B(buf, 1)
+      C3_BAR()
+      C3_MFMA(1, 1)
+      // phase 3: quadrant (0,1); stage B-half1 (unless staged at p1)
+      if constexpr (!B_BOTH) stage_b(nxt, 1, sk0);
+      C3_READ_A(buf, 0)
+      c3_waitbar<W_P3>();
+      C3_MFMA(0, 1)
+    }
+  
+    {  // peeled last tile
+      const int buf = (nkt - 1) & 1;
+      C3_READ_A(buf, 0)
+      C3_READ_B(buf, 0)
+      c3_waitbar<P_P0>();
+      C3_MFMA(0, 0)
+      C3_READ_A(buf, 1)
+      c3_waitbar<P_P1>();
+      C3_MFMA(1, 0)
+      C3_READ_B(buf, 1)
+      C3_BAR()
+      C3_MFMA(1, 1)
+      C3_READ_A(buf, 0)
+      C3_MFMA(0, 1)
+    }
   }
 
-  {  // peeled last tile
-    const int buf = (nkt - 1) & 1;
-    C3_READ_A(buf, 0)
-    C3_READ_B(buf, 0)
-    c3_waitbar<P_P0>();
-    C3_MFMA(0, 0)
-    C3_READ_A(buf, 1)
-    c3_waitbar<P_P1>();
-    C3_MFMA(1, 0)
-    C3_READ_B(buf, 1)
-    C3_BAR()
-    C3_MFMA(1, 1)
-    C3_READ_A(buf, 0)
-    C3_MFMA(0, 1)
-  }
-
+#undef C3_READ_B1
+#undef C3_MFMA1
 #undef C3_READ_A
 #undef C3_READ_B
 #undef C3_MFMA
