@@ -171,7 +171,8 @@ class _Conv1x1StatsFn(torch.autograd.Function):
             x, w = ctx.saved_tensors
             N, C, H, W = x.shape
             Cout = w.shape[0]
-            gy2 = gy.contiguous(memory_format=torch.channels_last)                 .permute(0, 2, 3, 1).reshape(-1, Cout)
+            gy = gy.contiguous(memory_format=torch.channels_last)
+            gy2 = gy.permute(0, 2, 3, 1).reshape(-1, Cout)
             w2 = w.reshape(Cout, C)
             gres2 = gres.permute(0, 2, 3, 1).reshape(-1, C)
             if gres2.is_contiguous():
@@ -184,9 +185,8 @@ class _Conv1x1StatsFn(torch.autograd.Function):
             else:
                 gx = res2.view(N, H, W, C).permute(0, 3, 1, 2)
             _, gw, _ = torch.ops.aten.convolution_backward(
-                gy.contiguous(memory_format=torch.channels_last), x, w,
-                None, [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
-                [False, True, False])
+                gy, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0],
+                1, [False, True, False])
             return gx, gw, None
         gx, gw = _Conv1x1Fn.backward(ctx, gy)
         return gx, gw, None
