@@ -67,6 +67,22 @@ torch::Tensor gemm_btw_acc_bf16(torch::Tensor a, torch::Tensor b,
                                 torch::Tensor out);
 torch::Tensor gemm_bt256p_acc_bf16(torch::Tensor a, torch::Tensor b,
                                    torch::Tensor out);
+torch::Tensor gemm_bt256p_bnbwd_bf16(
+    torch::Tensor a, torch::Tensor b, torch::Tensor bn_x,
+    torch::Tensor bn_mean, torch::Tensor bn_rstd, torch::Tensor bn_gamma,
+    torch::Tensor bn_beta, torch::Tensor bn_partials);
+torch::Tensor gemm_btw_bnbwd_bf16(torch::Tensor a, torch::Tensor b,
+                                  torch::Tensor bn_x,
+                                  torch::Tensor bn_mean,
+                                  torch::Tensor bn_rstd,
+                                  torch::Tensor bn_gamma,
+                                  torch::Tensor bn_beta,
+                                  torch::Tensor bn_partials);
+torch::Tensor conv3x3s1_fwd_bnbwd_bf16(
+    torch::Tensor x, torch::Tensor w, torch::Tensor stats,
+    torch::Tensor bn_x, torch::Tensor bn_mean, torch::Tensor bn_rstd,
+    torch::Tensor bn_gamma, torch::Tensor bn_beta,
+    torch::Tensor bn_partials);
 torch::Tensor maxpool3x3s2_bwd(torch::Tensor gy, torch::Tensor idx,
                                long IH, long IW);
 torch::Tensor bn_bwd_input(torch::Tensor x, torch::Tensor gy,
@@ -133,6 +149,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "narrow-N phased MFMA GEMM (BN 128/64): D = A @ B^T; optional "
         "fused BN-stats partials",
         py::arg("a"), py::arg("b"), py::arg("stats") = py::none());
+  m.def("gemm_bt256p_bnbwd_bf16", &gemm_bt256p_bnbwd_bf16,
+        "256-tile phased GEMM with the produced-gy BN-backward reduce "
+        "fused into the epilogue");
+  m.def("gemm_btw_bnbwd_bf16", &gemm_btw_bnbwd_bf16,
+        "narrow-N phased GEMM with fused BN-backward reduce");
+  m.def("conv3x3s1_fwd_bnbwd_bf16", &conv3x3s1_fwd_bnbwd_bf16,
+        "conv3x3 (dgrad form) with fused BN-backward reduce");
   m.def("gemm_btw_acc_bf16", &gemm_btw_acc_bf16,
         "narrow-N phased GEMM, accumulate mode: out += A @ B^T");
   m.def("gemm_bt256p_acc_bf16", &gemm_bt256p_acc_bf16,

@@ -190,3 +190,84 @@ __device__ __forceinline__ void stats_epilogue(
   }
 }
 
+
+// BN-BACKWARD reduce fused into a dgrad epilogue: the kernel that
+// PRODUCES gy for a BN(+ReLU) layer's backward also reduces
+// (sum g' | sum g'*xhat) with g' = gy * relu_mask(xhat*gamma+beta),
+// xhat = (x - mean) * rstd — the standalone bn_bwd_reduce read pass
+// over (x, gy) collapses to one x read here (gy is in registers).
+// Same wave->block->bucketed-atomics shape as stats_epilogue.
+struct BnBwdArgs {
+  const short* x;        // BN input, channels_last [M, C] bf16 bits
+  const float* mean;
+  const float* rstd;
+  const short* gamma;    // bf16 bits
+  const short* beta;
+  float* partials;       // [CMN_STATS_BUCKETS][2*C], pre-zeroed
+};
+
+template <int MF, int WAVES_M, int WAVES_N, typename ACC>
+__device__ __forceinline__ void bnbwd_epilogue(
+    const BnBwdArgs& bn, char* lds, const ACC& acc, int lane, int wave,
+    int wr, int wc, long row_base, int tile_n, int Cout, unsigned bid) {
+  constexpr int BN = 64 * WAVES_N;
+  const int fr = lane & 15;
+  const int col0 = tile_n * BN + wc * 64 + fr;
+  float mn[4], rs[4], gm[4], bt[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const int c = col0 + n * 16;
+    mn[n] = bn.mean[c];
+    rs[n] = bn.rstd[c];
+    gm[n] = float(reinterpret_cast<const c10::BFloat16*>(bn.gamma)[c]);
+    bt[n] = float(reinterpret_cast<const c10::BFloat16*>(bn.beta)[c]);
+  }
+  float s[4], qq[4];
+#pragma unroll
+  for (int n = 0; n < 4; ++n) { s[n] = 0.f; qq[n] = 0.f; }
+#pragma unroll
+  for (int m = 0; m < 2 * MF; ++m)
+#pragma unroll
+    for (int rg = 0; rg < 4; ++rg) {
+      const long row = row_base + m * 16 + rg;
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const float xv = float(reinterpret_cast<const c10::BFloat16*>(
+            bn.x)[row * Cout + col0 + n * 16]);
+        const float xh = (xv - mn[n]) * rs[n];
+        const bool mask = (xh * gm[n] + bt[n]) > 0.f;
+        const float g = mask ? float(c10::BFloat16(acc[m][n][rg])) : 0.f;
+        s[n] += g;
+        qq[n] += g * xh;
+      }
+    }
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    s[n] += __shfl_xor(s[n], 16);
+    s[n] += __shfl_xor(s[n], 32);
+    qq[n] += __shfl_xor(qq[n], 16);
+    qq[n] += __shfl_xor(qq[n], 32);
+  }
+  float* sl = reinterpret_cast<float*>(lds);
+  if (lane < 16) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      sl[wave * 128 + lane + n * 16] = s[n];
+      sl[wave * 128 + 64 + lane + n * 16] = qq[n];
+    }
+  }
+  __syncthreads();
+  if (wr == 0) {
+    float ss = 0.f, qs = 0.f;
+#pragma unroll
+    for (int k = 0; k < WAVES_M; ++k) {
+      const int w2 = k * WAVES_N + wc;
+      ss += sl[w2 * 128 + lane];
+      qs += sl[w2 * 128 + 64 + lane];
+    }
+    const int c = tile_n * BN + wc * 64 + lane;
+    const long base = (long)(bid & (CMN_STATS_BUCKETS - 1)) * 2 * Cout;
+    atomicAdd(&bn.partials[base + c], ss);
+    atomicAdd(&bn.partials[base + Cout + c], qs);
+  }
+}

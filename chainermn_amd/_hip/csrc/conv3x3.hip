@@ -68,7 +68,8 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
     const short* __restrict__ Wt,  // [Cout, 3, 3, Cin]
     const short* __restrict__ ZP,  // 128-byte zero page
     short* __restrict__ Y,         // [N, H, W, Cout]
-    float* __restrict__ SP,        // BN stats partials or nullptr
+    float* __restrict__ SP,        // fwd BN stats partials or nullptr
+    BnBwdArgs BNA,                 // fused BN-bwd reduce (dgrad) or nulls
     int HW, int H, int Wd, int Cin, int Cout) {
   constexpr int BM = 256;
   constexpr int BN = 64 * WAVES_N;
@@ -373,6 +374,11 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
     stats_epilogue<MF, WAVES_M, WAVES_N>(SP, lds, acc, lane, wave, wr,
                                          wc, tile_n, Cout,
                                          (unsigned)blockIdx.x);
+  } else if (BNA.partials != nullptr) {
+    __syncthreads();
+    bnbwd_epilogue<MF, WAVES_M, WAVES_N>(BNA, lds, acc, lane, wave, wr,
+                                         wc, row_base, tile_n, Cout,
+                                         (unsigned)blockIdx.x);
   }
 }
 
@@ -385,8 +391,25 @@ torch::Tensor& zero_page(const torch::Tensor& like) {
 
 }  // namespace
 
+torch::Tensor conv3x3s1_fwd_bnbwd_bf16(
+    torch::Tensor x, torch::Tensor w, torch::Tensor stats,
+    torch::Tensor bn_x, torch::Tensor bn_mean, torch::Tensor bn_rstd,
+    torch::Tensor bn_gamma, torch::Tensor bn_beta,
+    torch::Tensor bn_partials);
+
 torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
                                  torch::Tensor stats) {
+  return conv3x3s1_fwd_bnbwd_bf16(x, w, stats, {}, {}, {}, {}, {}, {});
+}
+
+// dgrad form with the BN-backward reduce fused into the epilogue: the
+// produced tensor is gy for the BN whose saved (x, mean, rstd, gamma,
+// beta) are given; (sum g' | sum g'*xhat) partials land in bn_partials.
+torch::Tensor conv3x3s1_fwd_bnbwd_bf16(
+    torch::Tensor x, torch::Tensor w, torch::Tensor stats,
+    torch::Tensor bn_x, torch::Tensor bn_mean, torch::Tensor bn_rstd,
+    torch::Tensor bn_gamma, torch::Tensor bn_beta,
+    torch::Tensor bn_partials) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16
               && w.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.dim() == 4
@@ -419,25 +442,41 @@ torch::Tensor conv3x3s1_fwd_bf16(torch::Tensor x, torch::Tensor w,
                 "stats must be zeroed float32 [16, 2*Cout]");
     sp = stats.data_ptr<float>();
   }
+  BnBwdArgs bna = {nullptr, nullptr, nullptr, nullptr, nullptr, nullptr};
+  if (bn_partials.defined() && bn_partials.numel() > 0) {
+    TORCH_CHECK(sp == nullptr, "stats and bn_partials are exclusive");
+    TORCH_CHECK(bn_partials.scalar_type() == at::kFloat
+                && bn_partials.numel() == 16 * 2 * (long)Cout
+                && bn_partials.is_contiguous());
+    TORCH_CHECK(bn_x.scalar_type() == at::kBFloat16
+                && bn_x.is_contiguous(at::MemoryFormat::ChannelsLast)
+                && bn_x.numel() == M * (long)Cout);
+    TORCH_CHECK(bn_mean.numel() == Cout && bn_rstd.numel() == Cout
+                && bn_gamma.numel() == Cout && bn_beta.numel() == Cout);
+    bna = {(const short*)bn_x.data_ptr(), bn_mean.data_ptr<float>(),
+           bn_rstd.data_ptr<float>(), (const short*)bn_gamma.data_ptr(),
+           (const short*)bn_beta.data_ptr(),
+           bn_partials.data_ptr<float>()};
+  }
   const int HW = H * Wd;
   if (Cout % 256 == 0) {
     const long grid = (M / 256) * (Cout / 256);
     hipLaunchKernelGGL((conv3x3_igemm_kernel<128, 2, 4>),
                        dim3((unsigned)grid), dim3(C3_THREADS), 0,
-                       stream.stream(), xp, wp, zp, yp, sp, HW, H, Wd,
-                       Cin, Cout);
+                       stream.stream(), xp, wp, zp, yp, sp, bna, HW, H,
+                       Wd, Cin, Cout);
   } else if (Cout % 128 == 0) {
     const long grid = (M / 256) * (Cout / 128);
     hipLaunchKernelGGL((conv3x3_igemm_kernel<64, 4, 2>),
                        dim3((unsigned)grid), dim3(C3_THREADS), 0,
-                       stream.stream(), xp, wp, zp, yp, sp, HW, H, Wd,
-                       Cin, Cout);
+                       stream.stream(), xp, wp, zp, yp, sp, bna, HW, H,
+                       Wd, Cin, Cout);
   } else {
     const long grid = (M / 256) * (Cout / 64);
     hipLaunchKernelGGL((conv3x3_igemm_kernel<32, 8, 1>),
                        dim3((unsigned)grid), dim3(C3_THREADS), 0,
-                       stream.stream(), xp, wp, zp, yp, sp, HW, H, Wd,
-                       Cin, Cout);
+                       stream.stream(), xp, wp, zp, yp, sp, bna, HW, H,
+                       Wd, Cin, Cout);
   }
   return y;
 }

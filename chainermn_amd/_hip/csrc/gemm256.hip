@@ -172,7 +172,8 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256_bf16_kernel(
 template <int SCHED, bool ACC = false>
 __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
-    short* __restrict__ D, float* __restrict__ SP, int M, int N, int K) {
+    short* __restrict__ D, float* __restrict__ SP, BnBwdArgs BNA,
+    int M, int N, int K) {
   __shared__ __attribute__((aligned(16))) char lds[2 * 2 * G2_BM * G2_BK * 2];
   auto ldsA = [&](int buf) -> char* {
     return lds + buf * 2 * G2_BM * G2_BK * 2;
@@ -372,6 +373,11 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
     __syncthreads();
     stats_epilogue<4, 2, 4>(SP, lds, acc, lane, wave, wr, wc, tile_n, N,
                             (unsigned)blockIdx.x);
+  } else if (BNA.partials != nullptr) {
+    __syncthreads();
+    bnbwd_epilogue<4, 2, 4>(BNA, lds, acc, lane, wave, wr, wc,
+                            (long)row_base, tile_n, N,
+                            (unsigned)blockIdx.x);
   }
 }
 
@@ -379,7 +385,10 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
 
 static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
                                     int phased_sched,
-                                    torch::Tensor stats = {}) {
+                                    torch::Tensor stats = {},
+                                    BnBwdArgs bna = {nullptr, nullptr,
+                                                     nullptr, nullptr,
+                                                     nullptr, nullptr}) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16
               && b.scalar_type() == at::kBFloat16);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
@@ -411,22 +420,22 @@ static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
     case 1:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<0>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         sp, M, N, K);
+                         sp, bna, M, N, K);
       break;
     case 2:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<1>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         sp, M, N, K);
+                         sp, bna, M, N, K);
       break;
     case 3:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<2>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         sp, M, N, K);
+                         sp, bna, M, N, K);
       break;
     default:
       hipLaunchKernelGGL(gemm_bt256p_bf16_kernel<3>, dim3(grid),
                          dim3(G2_THREADS), 0, stream.stream(), ap, bp, dp,
-                         sp, M, N, K);
+                         sp, bna, M, N, K);
   }
   return d;
 }
@@ -460,9 +469,33 @@ torch::Tensor gemm_bt256p_acc_bf16(torch::Tensor a, torch::Tensor b,
   TORCH_CHECK(M % G2_BM == 0 && N % G2_BN == 0 && K % G2_BK == 0);
   auto stream = c10::hip::getCurrentHIPStream();
   const int grid = (M / G2_BM) * (N / G2_BN);
+  BnBwdArgs bna0 = {nullptr, nullptr, nullptr, nullptr, nullptr,
+                    nullptr};
   hipLaunchKernelGGL((gemm_bt256p_bf16_kernel<3, true>), dim3(grid),
                      dim3(G2_THREADS), 0, stream.stream(),
                      (const short*)a.data_ptr(), (const short*)b.data_ptr(),
-                     (short*)out.data_ptr(), nullptr, M, N, K);
+                     (short*)out.data_ptr(), nullptr, bna0, M, N, K);
   return out;
+}
+
+// D = A @ B^T with the BN-backward reduce of the produced gy fused into
+// the epilogue (see conv3x3.hip's bnbwd entry for semantics).
+torch::Tensor gemm_bt256p_bnbwd_bf16(
+    torch::Tensor a, torch::Tensor b, torch::Tensor bn_x,
+    torch::Tensor bn_mean, torch::Tensor bn_rstd, torch::Tensor bn_gamma,
+    torch::Tensor bn_beta, torch::Tensor bn_partials) {
+  const long N = b.size(0);
+  TORCH_CHECK(bn_partials.scalar_type() == at::kFloat
+              && bn_partials.numel() == 16 * 2 * N
+              && bn_partials.is_contiguous());
+  TORCH_CHECK(bn_x.scalar_type() == at::kBFloat16
+              && bn_x.numel() == a.size(0) * N);
+  TORCH_CHECK(bn_mean.numel() == N && bn_rstd.numel() == N
+              && bn_gamma.numel() == N && bn_beta.numel() == N);
+  BnBwdArgs bna = {(const short*)bn_x.data_ptr(),
+                   bn_mean.data_ptr<float>(), bn_rstd.data_ptr<float>(),
+                   (const short*)bn_gamma.data_ptr(),
+                   (const short*)bn_beta.data_ptr(),
+                   bn_partials.data_ptr<float>()};
+  return launch_gemm256(a, b, 1 + 3, {}, bna);
 }

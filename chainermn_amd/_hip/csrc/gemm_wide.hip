@@ -48,7 +48,8 @@ __device__ __forceinline__ void gw_waitbar() {
 template <int WM, int WAVES_M, int WAVES_N, bool ACC = false>
 __global__ __launch_bounds__(GW_THREADS, 1) void gemm_btw_kernel(
     const short* __restrict__ A, const short* __restrict__ B,
-    short* __restrict__ D, float* __restrict__ SP, int M, int N, int K) {
+    short* __restrict__ D, float* __restrict__ SP, BnBwdArgs BNA,
+    int M, int N, int K) {
   constexpr int BM = 256;
   constexpr int BN = 64 * WAVES_N;
   constexpr int MF = WM / 32;
@@ -237,10 +238,23 @@ __global__ __launch_bounds__(GW_THREADS, 1) void gemm_btw_kernel(
     stats_epilogue<MF, WAVES_M, WAVES_N>(SP, lds, acc, lane, wave, wr,
                                          wc, tile_n, N,
                                          (unsigned)blockIdx.x);
+  } else if (BNA.partials != nullptr) {
+    __syncthreads();
+    bnbwd_epilogue<MF, WAVES_M, WAVES_N>(BNA, lds, acc, lane, wave, wr,
+                                         wc, row_base, tile_n, N,
+                                         (unsigned)blockIdx.x);
   }
 }
 
 }  // namespace
+
+torch::Tensor gemm_btw_bnbwd_bf16(torch::Tensor a, torch::Tensor b,
+                                  torch::Tensor bn_x,
+                                  torch::Tensor bn_mean,
+                                  torch::Tensor bn_rstd,
+                                  torch::Tensor bn_gamma,
+                                  torch::Tensor bn_beta,
+                                  torch::Tensor bn_partials);
 
 torch::Tensor gemm_btw_bf16(torch::Tensor a, torch::Tensor b,
                             torch::Tensor stats) {
@@ -264,16 +278,61 @@ torch::Tensor gemm_btw_bf16(torch::Tensor a, torch::Tensor b,
   const short* ap = (const short*)a.data_ptr();
   const short* bp = (const short*)b.data_ptr();
   short* dp = (short*)d.data_ptr();
+  BnBwdArgs bna = {nullptr, nullptr, nullptr, nullptr, nullptr, nullptr};
   if (N % 128 == 0) {
     const long grid = ((long)M / 256) * (N / 128);
     hipLaunchKernelGGL((gemm_btw_kernel<64, 4, 2>), dim3((unsigned)grid),
                        dim3(GW_THREADS), 0, stream.stream(), ap, bp, dp,
-                       sp, M, N, K);
+                       sp, bna, M, N, K);
   } else {
     const long grid = ((long)M / 256) * (N / 64);
     hipLaunchKernelGGL((gemm_btw_kernel<32, 8, 1>), dim3((unsigned)grid),
                        dim3(GW_THREADS), 0, stream.stream(), ap, bp, dp,
-                       sp, M, N, K);
+                       sp, bna, M, N, K);
+  }
+  return d;
+}
+
+torch::Tensor gemm_btw_bnbwd_bf16(torch::Tensor a, torch::Tensor b,
+                                  torch::Tensor bn_x,
+                                  torch::Tensor bn_mean,
+                                  torch::Tensor bn_rstd,
+                                  torch::Tensor bn_gamma,
+                                  torch::Tensor bn_beta,
+                                  torch::Tensor bn_partials) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16
+              && b.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(M % 256 == 0 && N % 64 == 0 && K % 64 == 0);
+  TORCH_CHECK(bn_partials.scalar_type() == at::kFloat
+              && bn_partials.numel() == 16 * 2 * (long)N
+              && bn_partials.is_contiguous());
+  TORCH_CHECK(bn_x.scalar_type() == at::kBFloat16
+              && bn_x.numel() == (long)M * N);
+  TORCH_CHECK(bn_mean.numel() == N && bn_rstd.numel() == N
+              && bn_gamma.numel() == N && bn_beta.numel() == N);
+  auto d = torch::empty({M, N}, a.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  const short* ap = (const short*)a.data_ptr();
+  const short* bp = (const short*)b.data_ptr();
+  short* dp = (short*)d.data_ptr();
+  BnBwdArgs bna = {(const short*)bn_x.data_ptr(),
+                   bn_mean.data_ptr<float>(), bn_rstd.data_ptr<float>(),
+                   (const short*)bn_gamma.data_ptr(),
+                   (const short*)bn_beta.data_ptr(),
+                   bn_partials.data_ptr<float>()};
+  if (N % 128 == 0) {
+    const long grid = ((long)M / 256) * (N / 128);
+    hipLaunchKernelGGL((gemm_btw_kernel<64, 4, 2>), dim3((unsigned)grid),
+                       dim3(GW_THREADS), 0, stream.stream(), ap, bp, dp,
+                       nullptr, bna, M, N, K);
+  } else {
+    const long grid = ((long)M / 256) * (N / 64);
+    hipLaunchKernelGGL((gemm_btw_kernel<32, 8, 1>), dim3((unsigned)grid),
+                       dim3(GW_THREADS), 0, stream.stream(), ap, bp, dp,
+                       nullptr, bna, M, N, K);
   }
   return d;
 }
@@ -297,12 +356,18 @@ torch::Tensor gemm_btw_acc_bf16(torch::Tensor a, torch::Tensor b,
     const long grid = ((long)M / 256) * (N / 128);
     hipLaunchKernelGGL((gemm_btw_kernel<64, 4, 2, true>),
                        dim3((unsigned)grid), dim3(GW_THREADS), 0,
-                       stream.stream(), ap, bp, dp, nullptr, M, N, K);
+                       stream.stream(), ap, bp, dp, nullptr,
+                       BnBwdArgs{nullptr, nullptr, nullptr, nullptr,
+                                 nullptr, nullptr},
+                       M, N, K);
   } else {
     const long grid = ((long)M / 256) * (N / 64);
     hipLaunchKernelGGL((gemm_btw_kernel<32, 8, 1, true>),
                        dim3((unsigned)grid), dim3(GW_THREADS), 0,
-                       stream.stream(), ap, bp, dp, nullptr, M, N, K);
+                       stream.stream(), ap, bp, dp, nullptr,
+                       BnBwdArgs{nullptr, nullptr, nullptr, nullptr,
+                                 nullptr, nullptr},
+                       M, N, K);
   }
   return out;
 }

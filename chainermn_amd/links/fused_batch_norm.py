@@ -50,7 +50,7 @@ class _FusedBNAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, res, gamma, beta, eps, relu_mode, momentum,
                 r_mean, r_var, comm=None, unbias=None, pre_stats=None,
-                junction=None):
+                junction=None, bwd_handoff=None):
         C = x.shape[1]
         count = x.numel() // C
         world = comm.size if comm is not None else 1
@@ -117,6 +117,12 @@ class _FusedBNAct(torch.autograd.Function):
         ctx.comm = comm if world > 1 else None
         ctx.has_res = res is not None
         ctx.junction = junction
+        ctx.bwd_handoff = bwd_handoff
+        if bwd_handoff is not None and relu_mode == 1 and x.is_cuda:
+            # arm the handoff: the conv consuming this BN's output can
+            # fuse THIS BN's backward reduce into its dgrad epilogue
+            bwd_handoff.update(x=x, mean=mean, rstd=rstd,
+                               gamma=gamma_f, beta=beta_f)
         return y
 
     @staticmethod
@@ -133,10 +139,20 @@ class _FusedBNAct(torch.autograd.Function):
         C = x.shape[1]
         if x.is_cuda:
             ext = ops.require()
-            buf = torch.empty(2 * C, dtype=torch.float32, device=x.device)
-            ext.bn_bwd_reduce_act_into(
-                x, gy, mean, rstd, gamma_f, beta_f,
-                yout if yout is not None else _empty(x), relu_mode, buf)
+            handoff = getattr(ctx, 'bwd_handoff', None)
+            partials = handoff.pop('partials', None) \
+                if handoff is not None else None
+            if partials is not None:
+                # the dgrad kernel that produced gy already reduced
+                # (sum g' | sum g'*xhat) in its epilogue
+                buf = partials.sum(0)
+            else:
+                buf = torch.empty(2 * C, dtype=torch.float32,
+                                  device=x.device)
+                ext.bn_bwd_reduce_act_into(
+                    x, gy, mean, rstd, gamma_f, beta_f,
+                    yout if yout is not None else _empty(x), relu_mode,
+                    buf)
             if ctx.comm is not None:
                 import torch.distributed as dist
                 dist.all_reduce(buf, group=ctx.comm.group)
@@ -174,7 +190,8 @@ class _FusedBNAct(torch.autograd.Function):
             # may accumulate the main-path gradient into it in-kernel
             ctx.junction['gres'] = g_res
         return (gx, g_res, ggamma.to(gamma.dtype), gbeta.to(beta.dtype),
-                None, None, None, None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None, None,
+                None)
 
 
 class FusedBatchNorm2d(nn.Module):
@@ -236,12 +253,13 @@ class FusedBatchNorm2d(nn.Module):
         return self.running_mean, self.running_var, m
 
     def _train_forward(self, x, res, relu_mode, pre_stats=None,
-                       junction=None):
+                       junction=None, bwd_handoff=None):
         gamma, beta = self._gamma_beta(x)
         r_mean, r_var, m = self._running()
         y = _FusedBNAct.apply(x, res, gamma, beta, self.eps, relu_mode, m,
                               r_mean, r_var, self.comm,
-                              self._running_unbias, pre_stats, junction)
+                              self._running_unbias, pre_stats, junction,
+                              bwd_handoff)
         if self.track_running_stats:
             self.num_batches_tracked += 1
         return y
@@ -263,10 +281,11 @@ class FusedBatchNorm2d(nn.Module):
             y = torch.relu(y)
         return y.to(x.dtype)
 
-    def forward(self, x, pre_stats=None):
+    def forward(self, x, pre_stats=None, bwd_handoff=None):
         if self.training or not self.track_running_stats:
             return self._train_forward(x, None, self._relu_mode,
-                                       pre_stats)
+                                       pre_stats,
+                                       bwd_handoff=bwd_handoff)
         return self._eval_forward(x, None, self._relu_mode)
 
     def forward_add_relu(self, x, res, pre_stats=None, junction=None):

@@ -150,7 +150,7 @@ class _Conv1x1StatsFn(torch.autograd.Function):
     _ResidualEntry.backward then skips the engine's separate add."""
 
     @staticmethod
-    def forward(ctx, x, w, junction=None):
+    def forward(ctx, x, w, junction=None, bwd_bn=None):
         N, C, H, W = x.shape
         x2 = x.permute(0, 2, 3, 1).reshape(-1, C)
         w2 = w.reshape(w.shape[0], C)
@@ -158,6 +158,7 @@ class _Conv1x1StatsFn(torch.autograd.Function):
         y = y2.view(N, H, W, -1).permute(0, 3, 1, 2)
         ctx.save_for_backward(x, w)
         ctx.junction = junction
+        ctx.bwd_bn = bwd_bn
         if stats is None:
             stats = x.new_empty(0, dtype=torch.float32)
         ctx.mark_non_differentiable(stats)
@@ -165,6 +166,23 @@ class _Conv1x1StatsFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy, g_stats):
+        bnh = getattr(ctx, 'bwd_bn', None)
+        if bnh and 'x' in bnh:
+            # 1x1 dgrad with the downstream BN's backward reduce fused
+            x, w = ctx.saved_tensors
+            N, C, H, W = x.shape
+            Cout = w.shape[0]
+            gy = gy.contiguous(memory_format=torch.channels_last)
+            gy2 = gy.permute(0, 2, 3, 1).reshape(-1, Cout)
+            w2 = w.reshape(Cout, C)
+            gx2, partials = ops.gemm_nn_bnbwd(gy2, w2, bnh)
+            if partials is not None:
+                bnh['partials'] = partials
+            gx = gx2.view(N, H, W, C).permute(0, 3, 1, 2)
+            _, gw, _ = torch.ops.aten.convolution_backward(
+                gy, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0],
+                1, [False, True, False])
+            return gx, gw, None, None
         j = ctx.junction
         gres = j.pop('gres', None) if j is not None else None
         if gres is not None:
@@ -187,9 +205,9 @@ class _Conv1x1StatsFn(torch.autograd.Function):
             _, gw, _ = torch.ops.aten.convolution_backward(
                 gy, x, w, None, [1, 1], [0, 0], [1, 1], False, [0, 0],
                 1, [False, True, False])
-            return gx, gw, None
+            return gx, gw, None, None
         gx, gw = _Conv1x1Fn.backward(ctx, gy)
-        return gx, gw, None
+        return gx, gw, None, None
 
 
 class _ResidualEntry(torch.autograd.Function):
@@ -217,8 +235,9 @@ class _ResidualEntry(torch.autograd.Function):
 class _Conv3x3StatsFn(torch.autograd.Function):
 
     @staticmethod
-    def forward(ctx, x, w):
+    def forward(ctx, x, w, bwd_bn=None):
         ctx.save_for_backward(x, w)
+        ctx.bwd_bn = bwd_bn
         y, stats = ops.conv3x3_with_stats(x, w)
         if stats is None:
             stats = x.new_empty(0, dtype=torch.float32)
@@ -227,20 +246,31 @@ class _Conv3x3StatsFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy, g_stats):
+        bnh = getattr(ctx, 'bwd_bn', None)
+        if bnh and 'x' in bnh:
+            x, w = ctx.saved_tensors
+            gy = gy.contiguous(memory_format=torch.channels_last)
+            wt = w.flip(2, 3).permute(1, 0, 2, 3).contiguous(
+                memory_format=torch.channels_last)
+            gx, partials = ops.conv3x3_with_bnbwd(gy, wt, bnh)
+            if partials is not None:
+                bnh['partials'] = partials
+            gw = ops.conv3x3_wgrad(x, gy, w)
+            return gx, gw, None
         gx, gw = _Conv3x3Fn.backward(ctx, gy)
-        return gx, gw
+        return gx, gw, None
 
 
 def _stats_or_none(st):
     return st if st.numel() else None
 
 
-def _conv1x1_forward_stats(self, x, junction=None):
+def _conv1x1_forward_stats(self, x, junction=None, bwd_bn=None):
     """(y, BN-stats partials or None) — the fused conv->bn fast path."""
     if x.is_cuda and x.dim() == 4 \
             and x.is_contiguous(memory_format=torch.channels_last) \
             and not x.is_contiguous():
-        y, st = _Conv1x1StatsFn.apply(x, self.weight, junction)
+        y, st = _Conv1x1StatsFn.apply(x, self.weight, junction, bwd_bn)
         return y, _stats_or_none(st)
     return self.forward(x), None
 
@@ -248,13 +278,13 @@ def _conv1x1_forward_stats(self, x, junction=None):
 FusedConv1x1.forward_stats = _conv1x1_forward_stats
 
 
-def _conv3x3_forward_stats(self, x):
+def _conv3x3_forward_stats(self, x, bwd_bn=None):
     """(y, BN-stats partials or None) — the fused conv->bn fast path."""
     if (self.stride == 1 and x.is_cuda and x.dim() == 4
             and x.dtype == torch.bfloat16
             and x.is_contiguous(memory_format=torch.channels_last)
             and not x.is_contiguous()):
-        y, st = _Conv3x3StatsFn.apply(x, self.weight)
+        y, st = _Conv3x3StatsFn.apply(x, self.weight, bwd_bn)
         return y, _stats_or_none(st)
     return self.forward(x), None
 

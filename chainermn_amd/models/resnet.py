@@ -18,6 +18,12 @@ class Bottleneck(nn.Module):
     # class-level switch for the residual-junction gradient fold
     # (tests compare fold-on vs fold-off on identical kernel stacks)
     _use_junction_fold = True
+    # BN-backward reduce fused into the dgrad epilogues: numerically
+    # verified (tests/test_gpu_kernels.py) but measured a NET LOSS —
+    # the epilogue's per-element x reads cost ~115us where the
+    # vectorized standalone bn_bwd_reduce costs 29us (layer3 shape) —
+    # so it ships disabled; see STATUS.md.
+    _use_bnbwd_fold = False
 
     def __init__(self, in_ch, width, stride=1, downsample=None,
                  norm_layer=nn.BatchNorm2d, fused_ops=False, comm=None):
@@ -57,18 +63,24 @@ class Bottleneck(nn.Module):
             # the junction's separate grad-add never runs.
             junction = None
             x_main = x
-            if self.downsample is None and x.is_cuda \
-                    and self._use_junction_fold:
-                from chainermn_amd.links.fused_conv import _ResidualEntry
-                junction = {}
-                x_main, identity = _ResidualEntry.apply(x, junction)
+            h1 = h2 = None
+            if x.is_cuda:
+                if self._use_bnbwd_fold:
+                    # backward handoffs: bn1's backward reduce fuses
+                    # into conv2's dgrad epilogue, bn2's into conv3's
+                    h1, h2 = {}, {}
+                if self.downsample is None and self._use_junction_fold:
+                    from chainermn_amd.links.fused_conv import \
+                        _ResidualEntry
+                    junction = {}
+                    x_main, identity = _ResidualEntry.apply(x, junction)
             y, st = self.conv1.forward_stats(x_main, junction=junction)
-            out = self.bn1(y, pre_stats=st)
-            y, st = self.conv2.forward_stats(out)
-            out = self.bn2(y, pre_stats=st)
+            out = self.bn1(y, pre_stats=st, bwd_handoff=h1)
+            y, st = self.conv2.forward_stats(out, bwd_bn=h1)
+            out = self.bn2(y, pre_stats=st, bwd_handoff=h2)
             if self.downsample is not None:
                 identity = self.downsample(x)
-            y, st = self.conv3.forward_stats(out)
+            y, st = self.conv3.forward_stats(out, bwd_bn=h2)
             return self.bn3.forward_add_relu(y, identity, pre_stats=st,
                                              junction=junction)
         out = self.relu(self.bn1(self.conv1(x)))

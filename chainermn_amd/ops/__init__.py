@@ -467,3 +467,65 @@ def gemm_nn_accum(a, b, out):
     # native beta=1 epilogue — still one pass, no separate add kernel
     out.addmm_(a, b)
     return out, True
+
+
+def _bnbwd_tensors(bnh):
+    return (bnh['x'], bnh['mean'], bnh['rstd'], bnh['gamma'], bnh['beta'])
+
+
+def conv3x3_with_bnbwd(x, w, bnh):
+    """conv3x3 in its dgrad role, fusing the downstream BN(+ReLU)
+    backward reduce into the epilogue (partials from the handoff's saved
+    statistics).  Returns (y, partials|None)."""
+    import torch.nn.functional as F
+    eligible = (x.is_cuda and x.dtype == torch.bfloat16
+                and w.dtype == torch.bfloat16 and x.dim() == 4
+                and x.shape[1] % 64 == 0 and w.shape[0] % 64 == 0
+                and (x.shape[0] * x.shape[2] * x.shape[3]) % 256 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and w.is_contiguous(memory_format=torch.channels_last))
+    if not eligible:
+        return F.conv2d(x, w, padding=1), None
+    key = ('c3', tuple(x.shape), w.shape[0])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        ext = require()
+        choice = _decide_multi(key, {
+            'mfma': lambda: ext.conv3x3s1_fwd_bf16(x, w),
+            'miopen': lambda: F.conv2d(x, w, padding=1)})
+    if choice == 'mfma':
+        bx, mean, rstd, gamma, beta = _bnbwd_tensors(bnh)
+        partials = torch.zeros(16, 2 * w.shape[0], dtype=torch.float32,
+                               device=x.device)
+        y = require().conv3x3s1_fwd_bnbwd_bf16(
+            x, w, x.new_empty(0, dtype=torch.float32), bx, mean, rstd,
+            gamma, beta, partials)
+        return y, partials
+    return F.conv2d(x, w, padding=1), None
+
+
+def gemm_nn_bnbwd(a, b, bnh):
+    """gemm_nn in its 1x1-dgrad role with the downstream BN backward
+    reduce fused (see conv3x3_with_bnbwd).  Returns (d, partials|None)."""
+    eligible = (a.is_cuda and a.dtype == torch.bfloat16
+                and a.shape[0] % 256 == 0 and b.shape[1] % 64 == 0
+                and a.shape[1] % 64 == 0 and a.is_contiguous())
+    if not eligible:
+        return gemm_nn(a, b), None
+    key = ('nn', a.shape[0], b.shape[1], a.shape[1])
+    choice = _GEMM_CHOICE.get(key)
+    if choice is None:
+        return gemm_nn(a, b), None
+    if choice not in ('mfma256', 'mfmaw'):
+        return gemm_nn(a, b), None
+    bx, mean, rstd, gamma, beta = _bnbwd_tensors(bnh)
+    partials = torch.zeros(16, 2 * b.shape[1], dtype=torch.float32,
+                           device=a.device)
+    bt = b.t().contiguous()
+    if choice == 'mfma256' and b.shape[1] % 256 == 0:
+        d = require().gemm_bt256p_bnbwd_bf16(a, bt, bx, mean, rstd,
+                                             gamma, beta, partials)
+    else:
+        d = require().gemm_btw_bnbwd_bf16(a, bt, bx, mean, rstd, gamma,
+                                          beta, partials)
+    return d, partials

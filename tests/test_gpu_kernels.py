@@ -941,3 +941,54 @@ def test_residual_junction_fold_matches_unfused():
             ref.zero_grad()
     finally:
         Bottleneck._use_junction_fold = True
+
+
+def test_bnbwd_epilogue_matches_standalone_reduce():
+    """The dgrad-fused BN-backward reduce == bn_bwd_reduce_act_into
+    (relu_mode 1) computed on the produced gy."""
+    torch.manual_seed(91)
+    # GEMM form (1x1 dgrad): d = a @ b^T is the BN's gy
+    for M, N, K in [(512, 256, 128), (256, 128, 64), (512, 64, 192)]:
+        a = torch.randn(M, K, device=DEV, dtype=torch.bfloat16) * 0.1
+        b = torch.randn(N, K, device=DEV, dtype=torch.bfloat16)
+        bn_x = torch.randn(M, N, device=DEV, dtype=torch.bfloat16)
+        mean = torch.randn(N, device=DEV) * 0.1
+        rstd = torch.rand(N, device=DEV) + 0.5
+        gamma = torch.randn(N, device=DEV, dtype=torch.bfloat16)
+        beta = torch.randn(N, device=DEV, dtype=torch.bfloat16) * 0.1
+        partials = torch.zeros(16, 2 * N, device=DEV)
+        if N % 256 == 0:
+            d = _ext().gemm_bt256p_bnbwd_bf16(a, b, bn_x, mean, rstd,
+                                              gamma, beta, partials)
+        else:
+            d = _ext().gemm_btw_bnbwd_bf16(a, b, bn_x, mean, rstd,
+                                           gamma, beta, partials)
+        ref = torch.empty(2 * N, device=DEV)
+        _ext().bn_bwd_reduce_act_into(bn_x, d, mean, rstd, gamma, beta,
+                                      bn_x.new_empty(0), 1, ref)
+        got = partials.sum(0)
+        assert torch.allclose(got, ref, rtol=1e-3, atol=5e-2), \
+            (M, N, K, (got - ref).abs().max())
+    # conv form (3x3 dgrad)
+    x = (torch.randn(2, 64, 16, 16, device=DEV, dtype=torch.bfloat16)
+         * 0.1).to(memory_format=torch.channels_last)
+    w = (torch.randn(128, 64, 3, 3, device=DEV, dtype=torch.bfloat16)
+         * 0.05).to(memory_format=torch.channels_last)
+    Cout = 128
+    Mt = 2 * 16 * 16
+    bn_x = torch.randn(2, Cout, 16, 16, device=DEV, dtype=torch.bfloat16
+                       ).to(memory_format=torch.channels_last)
+    mean = torch.randn(Cout, device=DEV) * 0.1
+    rstd = torch.rand(Cout, device=DEV) + 0.5
+    gamma = torch.randn(Cout, device=DEV, dtype=torch.bfloat16)
+    beta = torch.randn(Cout, device=DEV, dtype=torch.bfloat16) * 0.1
+    partials = torch.zeros(16, 2 * Cout, device=DEV)
+    y = _ext().conv3x3s1_fwd_bnbwd_bf16(
+        x, w, x.new_empty(0, dtype=torch.float32), bn_x, mean, rstd,
+        gamma, beta, partials)
+    ref = torch.empty(2 * Cout, device=DEV)
+    _ext().bn_bwd_reduce_act_into(bn_x, y, mean, rstd, gamma, beta,
+                                  bn_x.new_empty(0), 1, ref)
+    got = partials.sum(0)
+    assert torch.allclose(got, ref, rtol=1e-3, atol=5e-2), \
+        (got - ref).abs().max()
