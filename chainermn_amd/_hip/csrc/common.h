@@ -204,70 +204,75 @@ struct BnBwdArgs {
   const short* gamma;    // bf16 bits
   const short* beta;
   float* partials;       // [CMN_STATS_BUCKETS][2*C], pre-zeroed
+  const short* d;        // the produced gy tile base (set by the kernel)
 };
 
 template <int MF, int WAVES_M, int WAVES_N, typename ACC>
 __device__ __forceinline__ void bnbwd_epilogue(
     const BnBwdArgs& bn, char* lds, const ACC& acc, int lane, int wave,
     int wr, int wc, long row_base, int tile_n, int Cout, unsigned bid) {
+  // COALESCED tile sweep: instead of walking the accumulator fragments
+  // (whose lane layout forces per-element x reads), re-read the block's
+  // just-written D tile (L2-hot) alongside x with 16-byte loads and
+  // accumulate per-channel (sum g' | sum g'*xhat) via LDS atomics.
+  // acc/wr/wc unused in this variant; row_base/tile_n locate the tile.
+  (void)acc; (void)wr; (void)wc;
   constexpr int BN = 64 * WAVES_N;
-  const int fr = lane & 15;
-  const int col0 = tile_n * BN + wc * 64 + fr;
-  float mn[4], rs[4], gm[4], bt[4];
+  const int t = wave * 64 + lane;
+  // thread -> fixed channel octet; (512 / (BN/8)) threads share one
+  const int g = t % (BN / 8);
+  const int tr = t / (BN / 8);
+  constexpr int RPT = 256 / (512 / (BN / 8));  // rows per thread
+  const long row0 = row_base;          // tile_m * 256 (caller passes)
+  const int c0 = tile_n * BN + g * 8;
+  float mn[8], rs[8], gm[8], bt[8];
 #pragma unroll
-  for (int n = 0; n < 4; ++n) {
-    const int c = col0 + n * 16;
-    mn[n] = bn.mean[c];
-    rs[n] = bn.rstd[c];
-    gm[n] = float(reinterpret_cast<const c10::BFloat16*>(bn.gamma)[c]);
-    bt[n] = float(reinterpret_cast<const c10::BFloat16*>(bn.beta)[c]);
+  for (int k = 0; k < 8; ++k) {
+    mn[k] = bn.mean[c0 + k];
+    rs[k] = bn.rstd[c0 + k];
+    gm[k] = float(reinterpret_cast<const c10::BFloat16*>(bn.gamma)[c0 + k]);
+    bt[k] = float(reinterpret_cast<const c10::BFloat16*>(bn.beta)[c0 + k]);
   }
-  float s[4], qq[4];
+  float s[8], qq[8];
 #pragma unroll
-  for (int n = 0; n < 4; ++n) { s[n] = 0.f; qq[n] = 0.f; }
+  for (int k = 0; k < 8; ++k) { s[k] = 0.f; qq[k] = 0.f; }
+  for (int r = 0; r < RPT; ++r) {
+    const long row = row0 + tr * RPT + r;
+    const long off = row * Cout + c0;
+    const uint4 xv4 = *reinterpret_cast<const uint4*>(
+        reinterpret_cast<const short*>(bn.x) + off);
+    const uint4 dv4 = *reinterpret_cast<const uint4*>(bn.d + off);
+    const unsigned* xu = reinterpret_cast<const unsigned*>(&xv4);
+    const unsigned* du = reinterpret_cast<const unsigned*>(&dv4);
 #pragma unroll
-  for (int m = 0; m < 2 * MF; ++m)
-#pragma unroll
-    for (int rg = 0; rg < 4; ++rg) {
-      const long row = row_base + m * 16 + rg;
-#pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        const float xv = float(reinterpret_cast<const c10::BFloat16*>(
-            bn.x)[row * Cout + col0 + n * 16]);
-        const float xh = (xv - mn[n]) * rs[n];
-        const bool mask = (xh * gm[n] + bt[n]) > 0.f;
-        const float g = mask ? float(c10::BFloat16(acc[m][n][rg])) : 0.f;
-        s[n] += g;
-        qq[n] += g * xh;
-      }
+    for (int k = 0; k < 8; ++k) {
+      const unsigned xbits = (k & 1) ? (xu[k >> 1] >> 16)
+                                     : (xu[k >> 1] & 0xFFFFu);
+      const unsigned dbits = (k & 1) ? (du[k >> 1] >> 16)
+                                     : (du[k >> 1] & 0xFFFFu);
+      c10::BFloat16 xb, db;
+      xb.x = (unsigned short)xbits;
+      db.x = (unsigned short)dbits;
+      const float xh = (float(xb) - mn[k]) * rs[k];
+      const bool mask = (xh * gm[k] + bt[k]) > 0.f;
+      const float gv = mask ? float(db) : 0.f;
+      s[k] += gv;
+      qq[k] += gv * xh;
     }
-#pragma unroll
-  for (int n = 0; n < 4; ++n) {
-    s[n] += __shfl_xor(s[n], 16);
-    s[n] += __shfl_xor(s[n], 32);
-    qq[n] += __shfl_xor(qq[n], 16);
-    qq[n] += __shfl_xor(qq[n], 32);
   }
-  float* sl = reinterpret_cast<float*>(lds);
-  if (lane < 16) {
+  // per-channel accumulation across the threads sharing an octet
+  float* sl = reinterpret_cast<float*>(lds);  // [2 * BN]
+  for (int i = t; i < 2 * BN; i += 512) sl[i] = 0.f;
+  __syncthreads();
 #pragma unroll
-    for (int n = 0; n < 4; ++n) {
-      sl[wave * 128 + lane + n * 16] = s[n];
-      sl[wave * 128 + 64 + lane + n * 16] = qq[n];
-    }
+  for (int k = 0; k < 8; ++k) {
+    atomicAdd(&sl[g * 8 + k], s[k]);
+    atomicAdd(&sl[BN + g * 8 + k], qq[k]);
   }
   __syncthreads();
-  if (wr == 0) {
-    float ss = 0.f, qs = 0.f;
-#pragma unroll
-    for (int k = 0; k < WAVES_M; ++k) {
-      const int w2 = k * WAVES_N + wc;
-      ss += sl[w2 * 128 + lane];
-      qs += sl[w2 * 128 + 64 + lane];
-    }
-    const int c = tile_n * BN + wc * 64 + lane;
+  if (t < BN) {
     const long base = (long)(bid & (CMN_STATS_BUCKETS - 1)) * 2 * Cout;
-    atomicAdd(&bn.partials[base + c], ss);
-    atomicAdd(&bn.partials[base + Cout + c], qs);
+    atomicAdd(&bn.partials[base + tile_n * BN + t], sl[t]);
+    atomicAdd(&bn.partials[base + Cout + tile_n * BN + t], sl[BN + t]);
   }
 }

@@ -377,8 +377,8 @@ __global__ __launch_bounds__(C3_THREADS, 1) void conv3x3_igemm_kernel(
   } else if (BNA.partials != nullptr) {
     __syncthreads();
     bnbwd_epilogue<MF, WAVES_M, WAVES_N>(BNA, lds, acc, lane, wave, wr,
-                                         wc, row_base, tile_n, Cout,
-                                         (unsigned)blockIdx.x);
+                                         wc, tile_m * 256L, tile_n,
+                                         Cout, (unsigned)blockIdx.x);
   }
 }
 
@@ -442,7 +442,8 @@ torch::Tensor conv3x3s1_fwd_bnbwd_bf16(
                 "stats must be zeroed float32 [16, 2*Cout]");
     sp = stats.data_ptr<float>();
   }
-  BnBwdArgs bna = {nullptr, nullptr, nullptr, nullptr, nullptr, nullptr};
+  BnBwdArgs bna = {nullptr, nullptr, nullptr, nullptr, nullptr,
+                   nullptr, nullptr};
   if (bn_partials.defined() && bn_partials.numel() > 0) {
     TORCH_CHECK(sp == nullptr, "stats and bn_partials are exclusive");
     TORCH_CHECK(bn_partials.scalar_type() == at::kFloat
@@ -456,8 +457,9 @@ torch::Tensor conv3x3s1_fwd_bnbwd_bf16(
     bna = {(const short*)bn_x.data_ptr(), bn_mean.data_ptr<float>(),
            bn_rstd.data_ptr<float>(), (const short*)bn_gamma.data_ptr(),
            (const short*)bn_beta.data_ptr(),
-           bn_partials.data_ptr<float>()};
+           bn_partials.data_ptr<float>(), nullptr};
   }
+  bna.d = yp;
   const int HW = H * Wd;
   if (Cout % 256 == 0) {
     const long grid = (M / 256) * (Cout / 256);

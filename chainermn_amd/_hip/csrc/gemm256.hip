@@ -376,7 +376,7 @@ __global__ __launch_bounds__(G2_THREADS, 1) void gemm_bt256p_bf16_kernel(
   } else if (BNA.partials != nullptr) {
     __syncthreads();
     bnbwd_epilogue<4, 2, 4>(BNA, lds, acc, lane, wave, wr, wc,
-                            (long)row_base, tile_n, N,
+                            (long)tile_m * 256, tile_n, N,
                             (unsigned)blockIdx.x);
   }
 }
@@ -388,7 +388,8 @@ static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
                                     torch::Tensor stats = {},
                                     BnBwdArgs bna = {nullptr, nullptr,
                                                      nullptr, nullptr,
-                                                     nullptr, nullptr}) {
+                                                     nullptr, nullptr,
+                                                     nullptr}) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16
               && b.scalar_type() == at::kBFloat16);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
@@ -402,6 +403,7 @@ static torch::Tensor launch_gemm256(torch::Tensor a, torch::Tensor b,
   const short* ap = (const short*)a.data_ptr();
   const short* bp = (const short*)b.data_ptr();
   short* dp = (short*)d.data_ptr();
+  bna.d = dp;
   float* sp = nullptr;
   if (stats.defined() && stats.numel() > 0) {
     TORCH_CHECK(phased_sched > 0, "stats need the phased kernel");
@@ -470,7 +472,7 @@ torch::Tensor gemm_bt256p_acc_bf16(torch::Tensor a, torch::Tensor b,
   auto stream = c10::hip::getCurrentHIPStream();
   const int grid = (M / G2_BM) * (N / G2_BN);
   BnBwdArgs bna0 = {nullptr, nullptr, nullptr, nullptr, nullptr,
-                    nullptr};
+                    nullptr, nullptr};
   hipLaunchKernelGGL((gemm_bt256p_bf16_kernel<3, true>), dim3(grid),
                      dim3(G2_THREADS), 0, stream.stream(),
                      (const short*)a.data_ptr(), (const short*)b.data_ptr(),
@@ -496,6 +498,6 @@ torch::Tensor gemm_bt256p_bnbwd_bf16(
                    bn_mean.data_ptr<float>(), bn_rstd.data_ptr<float>(),
                    (const short*)bn_gamma.data_ptr(),
                    (const short*)bn_beta.data_ptr(),
-                   bn_partials.data_ptr<float>()};
+                   bn_partials.data_ptr<float>(), nullptr};
   return launch_gemm256(a, b, 1 + 3, {}, bna);
 }

@@ -241,7 +241,7 @@ __global__ __launch_bounds__(GW_THREADS, 1) void gemm_btw_kernel(
   } else if (BNA.partials != nullptr) {
     __syncthreads();
     bnbwd_epilogue<MF, WAVES_M, WAVES_N>(BNA, lds, acc, lane, wave, wr,
-                                         wc, row_base, tile_n, N,
+                                         wc, tile_m * 256L, tile_n, N,
                                          (unsigned)blockIdx.x);
   }
 }
@@ -278,7 +278,8 @@ torch::Tensor gemm_btw_bf16(torch::Tensor a, torch::Tensor b,
   const short* ap = (const short*)a.data_ptr();
   const short* bp = (const short*)b.data_ptr();
   short* dp = (short*)d.data_ptr();
-  BnBwdArgs bna = {nullptr, nullptr, nullptr, nullptr, nullptr, nullptr};
+  BnBwdArgs bna = {nullptr, nullptr, nullptr, nullptr, nullptr,
+                   nullptr, nullptr};
   if (N % 128 == 0) {
     const long grid = ((long)M / 256) * (N / 128);
     hipLaunchKernelGGL((gemm_btw_kernel<64, 4, 2>), dim3((unsigned)grid),
@@ -322,7 +323,8 @@ torch::Tensor gemm_btw_bnbwd_bf16(torch::Tensor a, torch::Tensor b,
                    bn_mean.data_ptr<float>(), bn_rstd.data_ptr<float>(),
                    (const short*)bn_gamma.data_ptr(),
                    (const short*)bn_beta.data_ptr(),
-                   bn_partials.data_ptr<float>()};
+                   bn_partials.data_ptr<float>(), (const short*)nullptr};
+  bna.d = dp;
   if (N % 128 == 0) {
     const long grid = ((long)M / 256) * (N / 128);
     hipLaunchKernelGGL((gemm_btw_kernel<64, 4, 2>), dim3((unsigned)grid),
@@ -358,7 +360,7 @@ torch::Tensor gemm_btw_acc_bf16(torch::Tensor a, torch::Tensor b,
                        dim3((unsigned)grid), dim3(GW_THREADS), 0,
                        stream.stream(), ap, bp, dp, nullptr,
                        BnBwdArgs{nullptr, nullptr, nullptr, nullptr,
-                                 nullptr, nullptr},
+                                 nullptr, nullptr, nullptr},
                        M, N, K);
   } else {
     const long grid = ((long)M / 256) * (N / 64);
@@ -366,7 +368,7 @@ torch::Tensor gemm_btw_acc_bf16(torch::Tensor a, torch::Tensor b,
                        dim3((unsigned)grid), dim3(GW_THREADS), 0,
                        stream.stream(), ap, bp, dp, nullptr,
                        BnBwdArgs{nullptr, nullptr, nullptr, nullptr,
-                                 nullptr, nullptr},
+                                 nullptr, nullptr, nullptr},
                        M, N, K);
   }
   return out;

@@ -19,10 +19,10 @@ class Bottleneck(nn.Module):
     # (tests compare fold-on vs fold-off on identical kernel stacks)
     _use_junction_fold = True
     # BN-backward reduce fused into the dgrad epilogues: numerically
-    # verified (tests/test_gpu_kernels.py) but measured a NET LOSS —
-    # the epilogue's per-element x reads cost ~115us where the
-    # vectorized standalone bn_bwd_reduce costs 29us (layer3 shape) —
-    # so it ships disabled; see STATUS.md.
+    # verified (tests/test_gpu_kernels.py) but measured a NET LOSS even
+    # after vectorizing the epilogue sweep (+62us at the GEMM's
+    # 1-block/CU geometry vs the 29us standalone reduce at full
+    # occupancy, layer3 shape) — ships disabled; see STATUS.md.
     _use_bnbwd_fold = False
 
     def __init__(self, in_ch, width, stride=1, downsample=None,
