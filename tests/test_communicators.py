@@ -285,3 +285,34 @@ def _pure_rccl_split_worker(rank, size):
 
 def test_pure_rccl_split_preserves_class():
     run_distributed(_pure_rccl_split_worker, 4)
+
+
+def _world4_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    import torch.nn as nn
+    torch.manual_seed(0)
+    model = nn.Linear(10, 6)
+    for i, p in enumerate(model.parameters()):
+        p.grad = torch.full_like(p, float(rank + i))
+    comm.multi_node_mean_grad(model)
+    x = torch.full((8,), float(rank))
+    s = comm.allreduce(x)
+    return {'g0': next(model.parameters()).grad,
+            's': s, 'rank': rank, 'size': comm.size,
+            'split_color': comm.split(rank % 2, rank).rank}
+
+
+def test_world4_mean_grad_and_split():
+    """4-process world: mean-grad, allreduce and split (insurance for
+    the driver's 8-GPU run — everything before it ran at world 2)."""
+    results = run_distributed(_world4_worker, 4)
+    mean0 = sum(range(4)) / 4.0
+    for r in results:
+        g = torch.from_numpy(r['g0'])
+        assert torch.allclose(g, torch.full_like(g, mean0)), r['rank']
+        assert torch.from_numpy(r['s']).allclose(
+            torch.full((8,), 6.0))  # sum 0+1+2+3
+        assert r['size'] == 4
+        # split(color=rank%2): two sub-worlds of 2; sub-rank = index of
+        # this rank among same-color ranks
+        assert r['split_color'] == (r['rank'] // 2)

@@ -169,3 +169,27 @@ def test_fused_dp_preserves_channels_last():
     for p in model.parameters():
         if p.dim() == 4:
             assert p.data.is_contiguous(memory_format=torch.channels_last)
+
+
+def _world4_bucketed_worker(rank, size):
+    comm = chainermn_amd.create_communicator('naive')
+    model = _model(0)
+    opt = FusedDataParallelSGD(model, comm, lr=0.1, momentum=0.9,
+                               overlap='bucketed', bucket_mb=0.0001,
+                               grad_dtype=torch.float32)
+    for step in range(2):
+        opt.zero_grad()
+        torch.manual_seed(70 + step * size + rank)
+        model(torch.randn(5, 6)).pow(2).mean().backward()
+        opt.step()
+    return [p.detach().clone() for p in model.parameters()]
+
+
+def test_bucketed_overlap_world4():
+    """Bucketed engine at world 4 with tiny buckets (many buckets, all
+    ranks must issue the same fixed collective order)."""
+    results = run_distributed(_world4_bucketed_worker, 4)
+    results = [[torch.from_numpy(a) for a in r] for r in results]
+    for other in results[1:]:
+        for a, b in zip(results[0], other):
+            assert torch.allclose(a, b, atol=1e-6)
