@@ -151,3 +151,29 @@ def test_fused_mnbn_relu_matches_global_batch():
                               x.grad[rank * B:(rank + 1) * B], atol=1e-5)
         assert torch.allclose(torch.from_numpy(r['ggamma']),
                               bn.weight.grad, atol=1e-4)
+
+
+def test_create_mnbn_model_converts_fused_bn():
+    """A model built with the FUSED local BN classes converts too (the
+    +ReLU variant keeps its fusion via MultiNodeBNReLU2d)."""
+    import torch.nn as nn
+    from chainermn_amd.links.fused_batch_norm import (FusedBatchNorm2d,
+                                                      FusedBNReLU2d)
+    from chainermn_amd.links.batch_normalization import (
+        MultiNodeBatchNormalization, MultiNodeBNReLU2d)
+    comm = chainermn_amd.create_communicator('dummy')
+    m = nn.Sequential(FusedBNReLU2d(8), FusedBatchNorm2d(4),
+                      nn.BatchNorm2d(6))
+    with torch.no_grad():
+        m[0].weight.mul_(2.0)
+        m[1].running_var.fill_(3.0)
+    mn = create_mnbn_model(m, comm)
+    assert isinstance(mn[0], MultiNodeBNReLU2d)
+    assert isinstance(mn[1], MultiNodeBatchNormalization)
+    assert not isinstance(mn[1], MultiNodeBNReLU2d)
+    assert isinstance(mn[2], MultiNodeBatchNormalization)
+    assert torch.allclose(mn[0].weight, m[0].weight)
+    assert torch.allclose(mn[1].running_var, m[1].running_var)
+    # double conversion is a no-op
+    mn2 = create_mnbn_model(mn, comm)
+    assert isinstance(mn2[0], MultiNodeBNReLU2d)
