@@ -166,6 +166,8 @@ class _Conv1x1StatsFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gy, g_stats):
+        # junction and bwd_bn are mutually exclusive by construction
+        # (Bottleneck arms junction on conv1, bwd_bn on conv3)
         bnh = getattr(ctx, 'bwd_bn', None)
         if bnh and 'x' in bnh:
             # 1x1 dgrad with the downstream BN's backward reduce fused
@@ -263,6 +265,11 @@ class _Conv3x3StatsFn(torch.autograd.Function):
 
 def _stats_or_none(st):
     return st if st.numel() else None
+
+
+# forward_stats is attached to the module classes here (rather than in
+# the class bodies above) so the Function variants it dispatches to can
+# be defined after the base classes they reuse.
 
 
 def _conv1x1_forward_stats(self, x, junction=None, bwd_bn=None):
