@@ -91,3 +91,25 @@ def test_conv_ops_cpu_fallbacks():
     # bnbwd fallback returns no partials off-GPU
     r2, parts = ops.gemm_nn_bnbwd(a, bmat, {})
     assert parts is None and torch.allclose(r2, a @ bmat, atol=1e-5)
+
+
+def test_residual_entry_function_semantics():
+    """_ResidualEntry: explicit fork whose backward adds the two path
+    grads, unless the junction was marked 'fused' (then the identity
+    grad is already folded into the main grad and must not double)."""
+    import torch
+    from chainermn_amd.links.fused_conv import _ResidualEntry
+    x = torch.randn(4, requires_grad=True)
+    j = {}
+    a, b = _ResidualEntry.apply(x, j)
+    (a * 2 + b * 3).sum().backward()
+    assert torch.allclose(x.grad, torch.full((4,), 5.0))
+    # fused flag short-circuits the add (g_main already contains g_id)
+    x2 = torch.randn(4, requires_grad=True)
+    j2 = {}
+    a2, b2 = _ResidualEntry.apply(x2, j2)
+    loss = (a2 * 2 + b2 * 3).sum()
+    j2['fused'] = True
+    loss.backward()
+    assert torch.allclose(x2.grad, torch.full((4,), 2.0))
+    assert 'fused' not in j2  # consumed
