@@ -113,3 +113,36 @@ def test_residual_entry_function_semantics():
     loss.backward()
     assert torch.allclose(x2.grad, torch.full((4,), 2.0))
     assert 'fused' not in j2  # consumed
+
+
+def test_wgrad_magic_divisor_formula():
+    """Python mirror of wg_magic() in _hip/csrc/wgrad3x3.hip: the
+    round-up magic (mul, shift) pair must give exact floor division for
+    every n <= n_max.  The kernel-side correctness is covered end to end
+    by the GPU wgrad test; this guards the FORMULA (shift search +
+    exactness bound) against future edits."""
+    import random
+    rng = random.Random(13)
+
+    def magic(d, n_max):
+        if d & (d - 1) == 0:  # power of two incl. 1
+            k = d.bit_length() - 1
+            return (1 << 31, k - 1)
+        for s in range(30, -1, -1):
+            m = ((1 << (32 + s)) + d - 1) // d
+            if m >= (1 << 32):
+                continue
+            e = m * d - (1 << (32 + s))
+            if e == 0 or (1 << (32 + s)) // e > n_max:
+                return (m, s)
+        raise AssertionError(d)
+
+    n_max = 2 ** 31 - 1
+    for d in [1, 2, 3, 5, 6, 7, 9, 14, 16, 28, 49, 56, 64, 100, 196,
+              784, 3136, 12544, 50176]:
+        m, s = magic(d, n_max)
+        samples = list(range(0, 3000)) + \
+            [rng.randrange(0, n_max) for _ in range(20000)] + \
+            [n_max, n_max - 1, d - 1, d, d + 1, 2 * d - 1]
+        for n in samples:
+            assert (n * m) >> (32 + s) == n // d, (d, n, m, s)
