@@ -74,6 +74,7 @@ __global__ void batched_copy_kernel(const long* __restrict__ ptrs,
       V8<DST>::store(flat_dst + base, r);
     } else {
       if (local >= size) continue;  // pad region
+      if (pdst == nullptr) continue;  // param with no grad bound
       V8<SRC>::load(flat_src + base, r);
 #pragma unroll
       for (int k = 0; k < 8; ++k) r[k] *= scale;
