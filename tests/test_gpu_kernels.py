@@ -992,3 +992,29 @@ def test_bnbwd_epilogue_matches_standalone_reduce():
     got = partials.sum(0)
     assert torch.allclose(got, ref, rtol=1e-3, atol=5e-2), \
         (got - ref).abs().max()
+
+
+def test_resnet_fused_ops_fp32_composition():
+    """fused_ops model at fp32: convs fall back to MIOpen while the BN
+    kernels run their fp32 paths — the mixed composition must train."""
+    import chainermn_amd
+    from chainermn_amd.models import ResNet50
+    from chainermn_amd.parallel import FusedDataParallelSGD
+    comm = chainermn_amd.create_communicator('dummy')
+    torch.manual_seed(21)
+    model = ResNet50(num_classes=10, fused_ops=True).to(
+        device=DEV, memory_format=torch.channels_last)  # fp32
+    opt = FusedDataParallelSGD(model, comm, lr=0.05, momentum=0.9,
+                               overlap=False, grad_dtype=torch.float32)
+    x = torch.randn(8, 3, 64, 64, device=DEV
+                    ).to(memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (8,), device=DEV)
+    losses = []
+    for _ in range(6):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert all(l == l for l in losses)  # finite
+    assert losses[-1] < losses[0], losses
