@@ -12,7 +12,9 @@
 //
 //   Cout >= 256: BN=256, waves 2(M)x4(N), per-wave 128x64  (16 MFMA/ph)
 //   Cout == 128: BN=128, waves 4x2,       per-wave  64x64  ( 8 MFMA/ph)
-//   Cout ==  64: BN= 64, waves 8x1,       per-wave  32x64  ( 4 MFMA/ph)
+//   Cout ==  64: BN= 64, waves 8x1,       per-wave  32x64; runs the
+//                2-phase loop (8 MFMA + 1 barrier per phase) — the
+//                4-phase schedule left this geometry wait-bound
 //
 // The A operand (im2col view of x) is staged by global_load_lds with the
 // PER-LANE SOURCE address doing the im2col: each lane's (n,oh,ow) base
